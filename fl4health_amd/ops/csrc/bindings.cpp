@@ -1,0 +1,196 @@
+// Python bindings for the CDNA4 flat-buffer FL kernels (see flat_ops.hip).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void launch_axpby(float*, const float*, float, float, int64_t, hipStream_t);
+void launch_prox_sgd(float*, const float*, const float*, float*, float, float, float, float, int,
+                     int64_t, hipStream_t);
+void launch_scaffold_sgd(float*, const float*, const float*, const float*, float, float, int64_t,
+                         hipStream_t);
+void launch_scaffold_variate(float*, float*, const float*, const float*, const float*, float,
+                             int64_t, hipStream_t);
+void launch_server_opt(float*, const float*, float*, float*, float*, int, float, float, float,
+                       float, float, int64_t, hipStream_t);
+void launch_reduce(const float*, const float*, double*, double*, int, int64_t, hipStream_t);
+void launch_clip_delta(float*, const float*, const float*, const double*, float, float*, int64_t,
+                       hipStream_t);
+void launch_gaussian_noise(float*, float, float, uint64_t, uint64_t, int64_t, hipStream_t);
+void launch_bernoulli_mask(const float*, const float*, float*, float*, uint64_t, uint64_t, int,
+                           int64_t, hipStream_t);
+void launch_per_sample_sqnorm(const float*, float*, int64_t, int64_t, hipStream_t);
+void launch_clip_rowsum(const float*, const float*, float*, float, int64_t, int64_t, hipStream_t);
+void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, int64_t,
+                      hipStream_t);
+void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
+}
+
+namespace {
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+}
+
+hipStream_t stream() { return at::cuda::getCurrentCUDAStream().stream(); }
+
+void axpby_(torch::Tensor y, torch::Tensor x, double a, double b) {
+  check_f32(y, "y");
+  check_f32(x, "x");
+  TORCH_CHECK(x.numel() == y.numel(), "size mismatch");
+  launch_axpby(y.data_ptr<float>(), x.data_ptr<float>(), (float)a, (float)b, y.numel(), stream());
+}
+
+void prox_sgd_step_(torch::Tensor p, torch::Tensor g, c10::optional<torch::Tensor> w0,
+                    c10::optional<torch::Tensor> mbuf, double lr, double mu, double momentum,
+                    double weight_decay, bool nesterov) {
+  check_f32(p, "p");
+  check_f32(g, "g");
+  const float* w0p = nullptr;
+  float* mp = nullptr;
+  if (w0.has_value()) { check_f32(*w0, "w0"); w0p = w0->data_ptr<float>(); }
+  if (mbuf.has_value()) { check_f32(*mbuf, "mbuf"); mp = mbuf->data_ptr<float>(); }
+  launch_prox_sgd(p.data_ptr<float>(), g.data_ptr<float>(), w0p, mp, (float)lr, (float)mu,
+                  (float)momentum, (float)weight_decay, nesterov ? 1 : 0, p.numel(), stream());
+}
+
+void scaffold_sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor c, torch::Tensor ci,
+                        double lr, double weight_decay) {
+  check_f32(p, "p"); check_f32(g, "g"); check_f32(c, "c"); check_f32(ci, "ci");
+  launch_scaffold_sgd(p.data_ptr<float>(), g.data_ptr<float>(), c.data_ptr<float>(),
+                      ci.data_ptr<float>(), (float)lr, (float)weight_decay, p.numel(), stream());
+}
+
+void scaffold_variate_update_(torch::Tensor ci, torch::Tensor dci, torch::Tensor c,
+                              torch::Tensor x_start, torch::Tensor y_end, double inv_klr) {
+  check_f32(ci, "ci"); check_f32(dci, "dci"); check_f32(c, "c");
+  check_f32(x_start, "x_start"); check_f32(y_end, "y_end");
+  launch_scaffold_variate(ci.data_ptr<float>(), dci.data_ptr<float>(), c.data_ptr<float>(),
+                          x_start.data_ptr<float>(), y_end.data_ptr<float>(), (float)inv_klr,
+                          ci.numel(), stream());
+}
+
+void server_opt_step_(torch::Tensor x, torch::Tensor delta, torch::Tensor m, torch::Tensor v,
+                      torch::Tensor dt, int64_t kind, double b1, double b2, double b3, double lr,
+                      double tau) {
+  check_f32(x, "x"); check_f32(delta, "delta"); check_f32(m, "m"); check_f32(v, "v");
+  check_f32(dt, "dt");
+  launch_server_opt(x.data_ptr<float>(), delta.data_ptr<float>(), m.data_ptr<float>(),
+                    v.data_ptr<float>(), dt.data_ptr<float>(), (int)kind, (float)b1, (float)b2,
+                    (float)b3, (float)lr, (float)tau, x.numel(), stream());
+}
+
+torch::Tensor reduce_op(torch::Tensor x, c10::optional<torch::Tensor> y, int64_t mode) {
+  check_f32(x, "x");
+  const float* yp = nullptr;
+  if (y.has_value()) {
+    check_f32(*y, "y");
+    TORCH_CHECK(y->numel() == x.numel(), "size mismatch");
+    yp = y->data_ptr<float>();
+  }
+  auto opts = torch::TensorOptions().dtype(torch::kFloat64).device(x.device());
+  auto partial = torch::empty({1024}, opts);
+  auto out = torch::empty({1}, opts);
+  launch_reduce(x.data_ptr<float>(), yp, partial.data_ptr<double>(), out.data_ptr<double>(),
+                (int)mode, x.numel(), stream());
+  return out;
+}
+
+torch::Tensor clip_delta(torch::Tensor w, torch::Tensor w0, torch::Tensor sqnorm,
+                         double clip_bound, c10::optional<torch::Tensor> clip_bit) {
+  check_f32(w, "w"); check_f32(w0, "w0");
+  TORCH_CHECK(sqnorm.scalar_type() == torch::kFloat64 && sqnorm.is_cuda(), "sqnorm f64 gpu");
+  auto out = torch::empty_like(w);
+  float* bitp = nullptr;
+  if (clip_bit.has_value()) { check_f32(*clip_bit, "clip_bit"); bitp = clip_bit->data_ptr<float>(); }
+  launch_clip_delta(out.data_ptr<float>(), w.data_ptr<float>(), w0.data_ptr<float>(),
+                    sqnorm.data_ptr<double>(), (float)clip_bound, bitp, w.numel(), stream());
+  return out;
+}
+
+void gaussian_noise_(torch::Tensor x, double a, double sigma, int64_t seed, int64_t offset) {
+  check_f32(x, "x");
+  launch_gaussian_noise(x.data_ptr<float>(), (float)a, (float)sigma, (uint64_t)seed,
+                        (uint64_t)offset, x.numel(), stream());
+}
+
+std::vector<torch::Tensor> bernoulli_mask(torch::Tensor scores, c10::optional<torch::Tensor> w,
+                                          int64_t seed, int64_t offset, bool apply_sigmoid) {
+  check_f32(scores, "scores");
+  auto mask = torch::empty_like(scores);
+  torch::Tensor weff;
+  const float* wp = nullptr;
+  float* weffp = nullptr;
+  if (w.has_value()) {
+    check_f32(*w, "w");
+    weff = torch::empty_like(*w);
+    wp = w->data_ptr<float>();
+    weffp = weff.data_ptr<float>();
+  }
+  launch_bernoulli_mask(scores.data_ptr<float>(), wp, mask.data_ptr<float>(), weffp,
+                        (uint64_t)seed, (uint64_t)offset, apply_sigmoid ? 1 : 0, scores.numel(),
+                        stream());
+  if (w.has_value()) return {mask, weff};
+  return {mask};
+}
+
+void per_sample_sqnorm_(torch::Tensor g, torch::Tensor out) {
+  check_f32(g, "g");
+  check_f32(out, "out");
+  int64_t B = g.size(0);
+  int64_t D = g.numel() / B;
+  TORCH_CHECK(out.numel() == B, "out must be [B]");
+  launch_per_sample_sqnorm(g.data_ptr<float>(), out.data_ptr<float>(), B, D, stream());
+}
+
+void clip_rowsum_(torch::Tensor g, torch::Tensor sqnorms, torch::Tensor out, double clip_bound) {
+  check_f32(g, "g"); check_f32(sqnorms, "sqnorms"); check_f32(out, "out");
+  int64_t B = g.size(0);
+  int64_t D = g.numel() / B;
+  TORCH_CHECK(out.numel() == D, "out must be [D]");
+  launch_clip_rowsum(g.data_ptr<float>(), sqnorms.data_ptr<float>(), out.data_ptr<float>(),
+                     (float)clip_bound, B, D, stream());
+}
+
+void confusion_counts_(torch::Tensor preds, torch::Tensor targets, torch::Tensor out) {
+  TORCH_CHECK(preds.is_cuda() && preds.scalar_type() == torch::kInt64 && preds.is_contiguous());
+  TORCH_CHECK(targets.is_cuda() && targets.scalar_type() == torch::kInt64 && targets.is_contiguous());
+  TORCH_CHECK(out.is_cuda() && out.scalar_type() == torch::kInt64 && out.is_contiguous());
+  int C = (int)out.size(0);
+  TORCH_CHECK(out.size(1) == 4, "out must be [C,4]");
+  launch_confusion(preds.data_ptr<int64_t>(), targets.data_ptr<int64_t>(),
+                   reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()), C,
+                   preds.numel(), stream());
+}
+
+torch::Tensor weighted_sum_rows(torch::Tensor stack, torch::Tensor w) {
+  check_f32(stack, "stack");
+  check_f32(w, "w");
+  int K = (int)stack.size(0);
+  int64_t n = stack.numel() / K;
+  TORCH_CHECK(w.numel() == K, "w must be [K]");
+  auto out = torch::empty({n}, stack.options());
+  launch_weighted_sum_rows(stack.data_ptr<float>(), w.data_ptr<float>(), out.data_ptr<float>(), K,
+                           n, stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("axpby_", &axpby_, "y = a*x + b*y (in-place)");
+  m.def("prox_sgd_step_", &prox_sgd_step_, "fused proximal SGD step");
+  m.def("scaffold_sgd_step_", &scaffold_sgd_step_, "fused SCAFFOLD-corrected SGD step");
+  m.def("scaffold_variate_update_", &scaffold_variate_update_, "SCAFFOLD control variate update");
+  m.def("server_opt_step_", &server_opt_step_, "fused FedOpt/Flash server step");
+  m.def("reduce_op", &reduce_op, "deterministic reduction (0 sqnorm,1 sqdiff,2 dot,3 sum)");
+  m.def("clip_delta", &clip_delta, "flat-clip weight delta");
+  m.def("gaussian_noise_", &gaussian_noise_, "x = a*x + sigma*N(0,1), philox");
+  m.def("bernoulli_mask", &bernoulli_mask, "bernoulli(sigmoid(scores)) mask (+masked weight)");
+  m.def("per_sample_sqnorm_", &per_sample_sqnorm_, "accumulate per-sample grad sq norms");
+  m.def("clip_rowsum_", &clip_rowsum_, "clipped per-sample grad sum");
+  m.def("confusion_counts_", &confusion_counts_, "streaming TP/FP/FN/TN counts");
+  m.def("weighted_sum_rows", &weighted_sum_rows, "out = sum_k w[k]*stack[k]");
+}

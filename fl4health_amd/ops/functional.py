@@ -1,0 +1,279 @@
+"""Functional op surface over the CDNA4 HIP kernels with CPU torch fallbacks.
+
+GPU path: `fl4health_amd._C` (in-tree HIP extension, gfx950). Missing extension
+on a GPU tensor is a hard error — no silent eager fallback on the device.
+CPU path: equivalent fp32 torch reference (test oracle; also used by the CPU
+test suite and gloo-backend multi-process tests).
+
+Reference behavior being reproduced (file:line cites into /root/reference):
+- prox_sgd_step_: fl4health/losses/weight_drift_loss.py:5-64 (mu/2*||w-w0||^2
+  penalty gradient) fused with torch.optim.SGD step.
+- scaffold_*: fl4health/clients/scaffold_client.py:137-197.
+- server_opt_step_: fl4health/strategies/flash.py:125-170 (+ FedOpt family).
+- clip_delta/gaussian_noise_: fl4health/clients/clipping_client.py:71-111 and
+  fl4health/strategies/noisy_aggregate.py:7-122.
+- bernoulli_mask: fl4health/utils/functions.py:10-42 (straight-through
+  Bernoulli sampling for FedPM masked layers).
+- confusion_counts_: fl4health/metrics/efficient_metrics_base.py:308-375.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+try:
+    from fl4health_amd import _C  # type: ignore[attr-defined]
+
+    HAS_EXT = True
+except ImportError:  # pragma: no cover - exercised only when ext missing
+    _C = None
+    HAS_EXT = False
+
+
+def _require_ext(op: str) -> None:
+    if not HAS_EXT:
+        raise RuntimeError(
+            f"fl4health_amd._C HIP extension is required for GPU op '{op}' but is not built. "
+            "Run: PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace"
+        )
+
+
+def axpby_(y: torch.Tensor, x: torch.Tensor, a: float, b: float) -> torch.Tensor:
+    """y = a*x + b*y in place."""
+    if y.is_cuda:
+        _require_ext("axpby_")
+        _C.axpby_(y, x, a, b)
+    else:
+        y.mul_(b).add_(x, alpha=a)
+    return y
+
+
+def prox_sgd_step_(
+    p: torch.Tensor,
+    g: torch.Tensor,
+    w0: torch.Tensor | None = None,
+    mbuf: torch.Tensor | None = None,
+    *,
+    lr: float,
+    mu: float = 0.0,
+    momentum: float = 0.0,
+    weight_decay: float = 0.0,
+    nesterov: bool = False,
+) -> None:
+    """Fused (proximal) SGD step: g' = g + wd*p + mu*(p - w0); momentum; p -= lr*u."""
+    if p.is_cuda:
+        _require_ext("prox_sgd_step_")
+        _C.prox_sgd_step_(p, g, w0, mbuf, lr, mu, momentum, weight_decay, nesterov)
+        return
+    geff = g.clone()
+    if weight_decay != 0.0:
+        geff.add_(p, alpha=weight_decay)
+    if w0 is not None and mu != 0.0:
+        geff.add_(p - w0, alpha=mu)
+    u = geff
+    if mbuf is not None:
+        mbuf.mul_(momentum).add_(geff)
+        u = geff.add(mbuf, alpha=momentum) if nesterov else mbuf
+    p.add_(u, alpha=-lr)
+
+
+def scaffold_sgd_step_(
+    p: torch.Tensor, g: torch.Tensor, c: torch.Tensor, ci: torch.Tensor, *, lr: float, weight_decay: float = 0.0
+) -> None:
+    """p -= lr * (g + c - ci) [+ weight decay]."""
+    if p.is_cuda:
+        _require_ext("scaffold_sgd_step_")
+        _C.scaffold_sgd_step_(p, g, c, ci, lr, weight_decay)
+        return
+    geff = g + c - ci
+    if weight_decay != 0.0:
+        geff.add_(p, alpha=weight_decay)
+    p.add_(geff, alpha=-lr)
+
+
+def scaffold_variate_update_(
+    ci: torch.Tensor,
+    dci: torch.Tensor,
+    c: torch.Tensor,
+    x_start: torch.Tensor,
+    y_end: torch.Tensor,
+    *,
+    inv_klr: float,
+) -> None:
+    """ci_new = ci - c + (x_start - y_end) * inv_klr; dci = ci_new - ci."""
+    if ci.is_cuda:
+        _require_ext("scaffold_variate_update_")
+        _C.scaffold_variate_update_(ci, dci, c, x_start, y_end, inv_klr)
+        return
+    ci_new = ci - c + (x_start - y_end) * inv_klr
+    dci.copy_(ci_new - ci)
+    ci.copy_(ci_new)
+
+
+SERVER_OPT_KINDS = {"fedavgm": 0, "fedadam": 1, "fedyogi": 2, "fedadagrad": 3, "flash": 4}
+
+
+def server_opt_step_(
+    x: torch.Tensor,
+    delta: torch.Tensor,
+    m: torch.Tensor,
+    v: torch.Tensor,
+    d: torch.Tensor,
+    *,
+    kind: str,
+    beta1: float = 0.9,
+    beta2: float = 0.99,
+    beta3: float = 0.9,
+    lr: float = 1.0,
+    tau: float = 1e-9,
+) -> None:
+    k = SERVER_OPT_KINDS[kind]
+    if x.is_cuda:
+        _require_ext("server_opt_step_")
+        _C.server_opt_step_(x, delta, m, v, d, k, beta1, beta2, beta3, lr, tau)
+        return
+    if k == 0:
+        m.mul_(beta1).add_(delta)
+        u = m
+    elif k == 3:
+        v.add_(delta * delta)
+        u = delta / (v.sqrt() + tau)
+    else:
+        m.mul_(beta1).add_(delta, alpha=1 - beta1)
+        d2 = delta * delta
+        if k == 1:
+            v.mul_(beta2).add_(d2, alpha=1 - beta2)
+            u = m / (v.sqrt() + tau)
+        elif k == 2:
+            v.sub_((1 - beta2) * torch.sign(v - d2) * d2)
+            u = m / (v.sqrt() + tau)
+        else:  # flash
+            vprev = v.clone()
+            v.mul_(beta2).add_(d2, alpha=1 - beta2)
+            d.mul_(beta3).add_(d2 - vprev, alpha=1 - beta3)
+            u = m / (v.sqrt() - d + tau)
+    x.add_(u, alpha=lr)
+
+
+def sq_norm(x: torch.Tensor) -> torch.Tensor:
+    """Deterministic ||x||^2 as 0-dim f64 tensor (device-resident on GPU)."""
+    if x.is_cuda:
+        _require_ext("sq_norm")
+        return _C.reduce_op(x, None, 0)[0]
+    return (x.double() * x.double()).sum()
+
+
+def sq_diff(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext("sq_diff")
+        return _C.reduce_op(x, y, 1)[0]
+    d = x.double() - y.double()
+    return (d * d).sum()
+
+
+def dot(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext("dot")
+        return _C.reduce_op(x, y, 2)[0]
+    return (x.double() * y.double()).sum()
+
+
+def clip_delta(
+    w: torch.Tensor, w0: torch.Tensor, clip_bound: float, return_bit: bool = True
+) -> tuple[torch.Tensor, torch.Tensor | None]:
+    """Clipped delta: min(1, C/||w-w0||)*(w-w0), plus clip-indicator bit."""
+    if w.is_cuda:
+        _require_ext("clip_delta")
+        sqn = _C.reduce_op(w, w0, 1)
+        bit = torch.zeros(1, dtype=torch.float32, device=w.device) if return_bit else None
+        out = _C.clip_delta(w, w0, sqn, clip_bound, bit)
+        return out, bit
+    delta = w - w0
+    nrm = float(delta.norm())
+    scale = clip_bound / nrm if (nrm > clip_bound and nrm > 0) else 1.0
+    bit = torch.tensor([1.0 if nrm <= clip_bound else 0.0]) if return_bit else None
+    return delta * scale, bit
+
+
+def gaussian_noise_(x: torch.Tensor, sigma: float, seed: int, offset: int = 0, a: float = 1.0) -> None:
+    """x = a*x + sigma*N(0,1) with counter-based Philox draws (replayable)."""
+    if x.is_cuda:
+        _require_ext("gaussian_noise_")
+        _C.gaussian_noise_(x, a, sigma, seed, offset)
+        return
+    gen = torch.Generator().manual_seed(seed + offset * 1000003)
+    noise = torch.randn(x.shape, generator=gen, dtype=x.dtype)
+    x.mul_(a).add_(noise, alpha=sigma)
+
+
+def bernoulli_mask(
+    scores: torch.Tensor,
+    w: torch.Tensor | None = None,
+    *,
+    seed: int,
+    offset: int = 0,
+    apply_sigmoid: bool = True,
+) -> tuple[torch.Tensor, torch.Tensor | None]:
+    """mask ~ Bernoulli(sigmoid(scores)); optionally also mask*w."""
+    if scores.is_cuda:
+        _require_ext("bernoulli_mask")
+        outs = _C.bernoulli_mask(scores, w, seed, offset, apply_sigmoid)
+        return (outs[0], outs[1]) if len(outs) == 2 else (outs[0], None)
+    probs = torch.sigmoid(scores) if apply_sigmoid else scores
+    gen = torch.Generator().manual_seed(seed + offset * 1000003)
+    mask = torch.bernoulli(probs, generator=gen)
+    return mask, (mask * w if w is not None else None)
+
+
+def per_sample_sqnorm_(g: torch.Tensor, out: torch.Tensor) -> None:
+    """out[b] += sum over non-batch dims of g[b]^2 (accumulated across layers)."""
+    if g.is_cuda:
+        _require_ext("per_sample_sqnorm_")
+        _C.per_sample_sqnorm_(g.contiguous(), out)
+        return
+    out.add_(g.reshape(g.shape[0], -1).pow(2).sum(dim=1))
+
+
+def clip_rowsum_(g: torch.Tensor, sqnorms: torch.Tensor, out: torch.Tensor, clip_bound: float) -> None:
+    """out[d] += sum_b min(1, C/(sqrt(sqnorms[b])+1e-6)) * g[b,d]."""
+    if g.is_cuda:
+        _require_ext("clip_rowsum_")
+        _C.clip_rowsum_(g.contiguous(), sqnorms, out.reshape(-1), clip_bound)
+        return
+    nrm = sqnorms.sqrt() + 1e-6
+    coef = torch.clamp(clip_bound / nrm, max=1.0)
+    out.reshape(-1).add_((coef.unsqueeze(1) * g.reshape(g.shape[0], -1)).sum(dim=0))
+
+
+def confusion_counts_(preds: torch.Tensor, targets: torch.Tensor, out: torch.Tensor) -> None:
+    """out[c] += (tp, fp, fn, tn) for class c from argmax preds/targets."""
+    if preds.is_cuda:
+        _require_ext("confusion_counts_")
+        _C.confusion_counts_(preds.contiguous(), targets.contiguous(), out)
+        return
+    n_classes = out.shape[0]
+    n = preds.numel()
+    for c in range(n_classes):
+        pc = preds == c
+        tc = targets == c
+        tp = int((pc & tc).sum())
+        fp = int(pc.sum()) - tp
+        fn = int(tc.sum()) - tp
+        out[c, 0] += tp
+        out[c, 1] += fp
+        out[c, 2] += fn
+        out[c, 3] += n - tp - fp - fn
+
+
+def weighted_sum_rows(stack: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """out = sum_k w[k] * stack[k] over [K, n] (deterministic fixed order)."""
+    if stack.is_cuda:
+        _require_ext("weighted_sum_rows")
+        return _C.weighted_sum_rows(stack, w)
+    return (w.unsqueeze(1) * stack).sum(dim=0)
+
+
+def noise_multiplier_sigma(noise_multiplier: float, clip_bound: float, num: int) -> float:
+    """Std-dev of per-coordinate Gaussian noise for a summed, clipped aggregate."""
+    return noise_multiplier * clip_bound / max(num, 1) * math.sqrt(1.0)

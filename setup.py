@@ -1,0 +1,33 @@
+"""Build the in-tree CDNA4 (gfx950) HIP extension for fl4health_amd.
+
+Usage:  PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+"""
+import os
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "fl4health_amd", "ops", "csrc")
+
+setup(
+    name="fl4health_amd",
+    version="0.1.0",
+    packages=["fl4health_amd"],
+    ext_modules=[
+        CUDAExtension(
+            name="fl4health_amd._C",
+            sources=[
+                os.path.join(CSRC, "bindings.cpp"),
+                os.path.join(CSRC, "flat_ops.hip"),
+            ],
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+)
