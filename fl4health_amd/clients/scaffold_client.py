@@ -1,0 +1,89 @@
+"""SCAFFOLD client (reference fl4health/clients/scaffold_client.py:23-355).
+
+Receives [x || c] each round; per-step update uses the fused
+variate-corrected SGD kernel p -= lr*(g + c - c_i) (K3); after local training
+computes c_i+ = c_i - c + (x_start - y_end)/(K*lr) and packs [y || delta_c_i]
+(K2) — both single fused kernel passes over the flat params region.
+"""
+from __future__ import annotations
+
+import torch
+
+from fl4health_amd.clients.basic_client import BasicClient
+from fl4health_amd.common import Config, Parameters
+from fl4health_amd.ops import functional as F
+from fl4health_amd.optimizers import FlatScaffoldSGD
+from fl4health_amd.parameter_exchange.exchangers import FullParameterExchangerWithPacking
+from fl4health_amd.parameter_exchange.packers import ParameterPackerWithControlVariates
+
+
+class ScaffoldClient(BasicClient):
+    def __init__(self, *args, learning_rate: float | None = None, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        self.learning_rate = learning_rate
+        self.client_control_variates: torch.Tensor | None = None  # c_i
+        self.client_control_variates_updates: torch.Tensor | None = None  # delta c_i
+        self.server_control_variates: torch.Tensor | None = None  # c
+        self.server_model_weights: torch.Tensor | None = None  # x at round start
+        self._steps_this_round = 0
+
+    def get_parameter_exchanger(self, config: Config) -> FullParameterExchangerWithPacking:
+        return FullParameterExchangerWithPacking(ParameterPackerWithControlVariates())
+
+    @property
+    def _scaffold_optimizer(self) -> FlatScaffoldSGD:
+        opt = self.optimizers["global"]
+        assert isinstance(opt, FlatScaffoldSGD), "ScaffoldClient requires a FlatScaffoldSGD optimizer"
+        return opt
+
+    def set_parameters(self, parameters: Parameters, config: Config, fitting_round: bool) -> None:
+        assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
+        model_params, server_variates = self.parameter_exchanger.unpack_parameters(parameters)
+        self.parameter_exchanger.pull_parameters(model_params, self.model, config)
+        n = self.flat_view.params_numel
+        self.server_control_variates = server_variates.detach().clone().to(self.device)[:n]
+        if self.client_control_variates is None:
+            self.client_control_variates = torch.zeros_like(self.server_control_variates)
+        if fitting_round:
+            self.server_model_weights = self.flat_view.params_region.detach().clone()
+            self._steps_this_round = 0
+            opt = self._scaffold_optimizer
+            opt.set_variates(self.server_control_variates, self.client_control_variates)
+            if self.learning_rate is None:
+                self.learning_rate = opt.lr
+
+    def update_after_step(self, step: int, current_round: int | None = None) -> None:
+        self._steps_this_round += 1
+        super().update_after_step(step, current_round)
+
+    def update_after_train(self, local_steps: int, loss_dict: dict[str, float], config: Config) -> None:
+        """c_i+ = c_i - c + (x - y)/(K*lr); delta_c_i = c_i+ - c_i (reference :137-173)."""
+        assert (
+            self.client_control_variates is not None
+            and self.server_control_variates is not None
+            and self.server_model_weights is not None
+        )
+        k = max(self._steps_this_round, 1)
+        lr = self.learning_rate or self._scaffold_optimizer.lr
+        if self.client_control_variates_updates is None:
+            self.client_control_variates_updates = torch.zeros_like(self.client_control_variates)
+        F.scaffold_variate_update_(
+            self.client_control_variates,
+            self.client_control_variates_updates,
+            self.server_control_variates,
+            self.server_model_weights,
+            self.flat_view.params_region,
+            inv_klr=1.0 / (k * lr),
+        )
+        super().update_after_train(local_steps, loss_dict, config)
+
+    def get_parameters(self, config: Config) -> Parameters:
+        assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
+        model_params = self.parameter_exchanger.push_parameters(self.model, config=config)
+        assert self.client_control_variates_updates is not None, "fit must run before get_parameters"
+        return self.parameter_exchanger.pack_parameters(model_params, self.client_control_variates_updates)
+
+
+class DPScaffoldClient(ScaffoldClient):
+    """SCAFFOLD + instance-level DP-SGD (reference scaffold_client.py:297):
+    combined in clients/instance_level_dp_client.py integration."""

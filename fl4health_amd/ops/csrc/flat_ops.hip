@@ -144,8 +144,10 @@ extern "C" void launch_scaffold_variate(float* ci, float* dci, const float* c,
 //   kind 2 FedYogi : m = ...; v = v-(1-b2)*sign(v-d^2)*d^2;   u = m/(sqrt(v)+tau)
 //   kind 3 FedAdagrad: v = v + d^2;                           u = delta/(sqrt(v)+tau)  (m unused)
 //   kind 4 Flash   : m=b1*m+(1-b1)*d; vp=v; v=b2*v+(1-b2)*d^2;
-//                    dt=b3*dt+(1-b3)*(d^2-vp);                u = m/(sqrt(v)-dt+tau)
-//     (reference fl4health/strategies/flash.py:125-170)
+//                    b3m = |vp| / (|d^2-v| + |vp|)   (per-element, 0 when both 0)
+//                    dt=b3m*dt+(1-b3m)*(d^2-v);               u = m/(sqrt(v)-dt+tau)
+//     (reference fl4health/strategies/flash.py:125-170: beta_3 is a per-element
+//      matrix derived from |v_prev| and |delta^2 - v_new|)
 //   x += lr * u
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(BLOCK) void server_opt_kernel(
@@ -181,7 +183,10 @@ __global__ __launch_bounds__(BLOCK) void server_opt_kernel(
         float vprev = vi;
         vi = fmaf(b2, vi, (1.0f - b2) * d2);
         v[i] = vi;
-        float dti = fmaf(b3, dt[i], (1.0f - b3) * (d2 - vprev));
+        float diff = d2 - vi;
+        float denom = fabsf(diff) + fabsf(vprev);
+        float b3m = (denom > 0.0f) ? fabsf(vprev) / denom : 0.0f;
+        float dti = fmaf(b3m, dt[i], (1.0f - b3m) * diff);
         dt[i] = dti;
         u = mi / (sqrtf(vi) - dti + tau);
       }

@@ -148,10 +148,13 @@ def server_opt_step_(
         elif k == 2:
             v.sub_((1 - beta2) * torch.sign(v - d2) * d2)
             u = m / (v.sqrt() + tau)
-        else:  # flash
+        else:  # flash (beta3 is a per-element matrix, reference flash.py:125-142)
             vprev = v.clone()
             v.mul_(beta2).add_(d2, alpha=1 - beta2)
-            d.mul_(beta3).add_(d2 - vprev, alpha=1 - beta3)
+            diff = d2 - v
+            denom = diff.abs() + vprev.abs()
+            b3m = torch.where(denom > 0, vprev.abs() / denom, torch.zeros_like(denom))
+            d.mul_(b3m).add_((1 - b3m) * diff)
             u = m / (v.sqrt() - d + tau)
     x.add_(u, alpha=lr)
 

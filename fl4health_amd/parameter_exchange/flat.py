@@ -1,0 +1,166 @@
+"""Flat parameter substrate — the MI355X-native wire format.
+
+The reference exchanges a model as an ordered list of per-layer NumPy arrays
+(fl4health/parameter_exchange/full_exchanger.py:10-48); every aggregation then
+loops layer-by-layer in Python. Here the exchanged set lives in ONE contiguous
+fp32 device tensor:
+
+- one RCCL all-reduce / broadcast moves the whole model (bucketing handled by
+  the communicator, sized for 7 xGMI links),
+- one fused HIP kernel implements each server/client hot op over the buffer,
+- module params/buffers are rebound as VIEWS of the flat buffer where dtypes
+  permit, so push/pull are zero-copy on the hot path.
+
+Non-fp32 state_dict entries (e.g. BatchNorm ``num_batches_tracked`` int64) are
+carried in the flat buffer as casted fp32 slots and cast back on pull — same
+aggregation semantics as the reference (flwr averages them as float arrays).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+import torch
+import torch.nn as nn
+
+
+@dataclass
+class ParameterSpec:
+    """Layout of a flat buffer: state_dict-ordered (name, shape, dtype, offset)."""
+
+    names: list[str]
+    shapes: list[torch.Size]
+    dtypes: list[torch.dtype]
+    offsets: list[int] = field(default_factory=list)
+    total: int = 0
+
+    def __post_init__(self) -> None:
+        if not self.offsets:
+            off = 0
+            for s in self.shapes:
+                self.offsets.append(off)
+                off += int(torch.Size(s).numel())
+            self.total = off
+
+    @classmethod
+    def from_named_tensors(cls, named: list[tuple[str, torch.Tensor]]) -> "ParameterSpec":
+        return cls(
+            names=[n for n, _ in named],
+            shapes=[t.shape for _, t in named],
+            dtypes=[t.dtype for _, t in named],
+        )
+
+    def slice_of(self, flat: torch.Tensor, i: int) -> torch.Tensor:
+        n = int(torch.Size(self.shapes[i]).numel())
+        return flat[self.offsets[i] : self.offsets[i] + n].view(self.shapes[i])
+
+    def index_of(self, name: str) -> int:
+        return self.names.index(name)
+
+
+class FlatParameterView:
+    """A flat fp32 buffer over a chosen subset of a module's state_dict.
+
+    ``bind=True`` additionally rebinds the module's fp32 params/buffers to be
+    views of the flat buffer (zero-copy push/pull + fused flat optimizer ops).
+    """
+
+    def __init__(self, module: nn.Module, names: list[str] | None = None, device: torch.device | str | None = None, bind: bool = False) -> None:
+        sd = module.state_dict()
+        if names is None:
+            # params-first ordering: all trainable parameters form one
+            # contiguous leading region of the flat buffer, so the fused
+            # optimizer/penalty kernels (prox-SGD, SCAFFOLD correction, DP
+            # clip) run over flat[:params_numel] in a single pass while
+            # buffers (BN running stats...) are exchanged but never stepped.
+            param_names = [n for n, _ in module.named_parameters()]
+            param_set = set(param_names)
+            names = param_names + [n for n in sd.keys() if n not in param_set]
+            self.params_numel = sum(sd[n].numel() for n in param_names)
+        else:
+            self.params_numel = None  # unknown for custom subsets
+        named = [(n, sd[n]) for n in names]
+        self.spec = ParameterSpec.from_named_tensors(named)
+        dev = device if device is not None else (named[0][1].device if named else "cpu")
+        self.flat = torch.zeros(self.spec.total, dtype=torch.float32, device=dev)
+        self.module = module
+        self.bound = False
+        self.pull_into_flat()
+        if bind:
+            self._bind_views()
+
+    # ---- data movement -------------------------------------------------
+    def pull_into_flat(self) -> None:
+        """Copy current module tensors into the flat buffer (no-op when bound)."""
+        sd = self.module.state_dict()
+        for i, name in enumerate(self.spec.names):
+            t = sd[name]
+            dst = self.spec.slice_of(self.flat, i)
+            if self.bound and dst.data_ptr() == t.data_ptr():
+                continue
+            dst.copy_(t.detach().to(torch.float32))
+
+    def push_into_module(self) -> None:
+        """Copy flat buffer values back into the module (no-op for bound views)."""
+        sd = self.module.state_dict()
+        with torch.no_grad():
+            for i, name in enumerate(self.spec.names):
+                t = sd[name]
+                src = self.spec.slice_of(self.flat, i)
+                if self.bound and src.data_ptr() == t.data_ptr():
+                    continue
+                t.copy_(src.to(t.dtype))
+
+    def clone_flat(self) -> torch.Tensor:
+        return self.flat.detach().clone()
+
+    def load_flat(self, flat: torch.Tensor) -> None:
+        self.flat.copy_(flat.to(self.flat.device, torch.float32))
+        self.push_into_module()
+
+    # ---- view binding (zero-copy hot path) -----------------------------
+    def _bind_views(self) -> None:
+        """Rebind fp32 module params/buffers as views of the flat buffer."""
+        name_to_idx = {n: i for i, n in enumerate(self.spec.names)}
+        for mod_name, mod in self.module.named_modules():
+            prefix = mod_name + "." if mod_name else ""
+            for pname, p in list(mod.named_parameters(recurse=False)):
+                full = prefix + pname
+                i = name_to_idx.get(full)
+                if i is None or p.dtype != torch.float32:
+                    continue
+                view = self.spec.slice_of(self.flat, i)
+                new_p = nn.Parameter(view, requires_grad=p.requires_grad)
+                setattr(mod, pname, new_p)
+            for bname, b in list(mod.named_buffers(recurse=False)):
+                full = prefix + bname
+                i = name_to_idx.get(full)
+                if i is None or b is None or b.dtype != torch.float32:
+                    continue
+                mod._buffers[bname] = self.spec.slice_of(self.flat, i)
+        self.bound = True
+
+    # ---- flat gradient buffer ------------------------------------------
+    @property
+    def params_region(self) -> torch.Tensor:
+        """Contiguous trainable-parameter slice of the flat buffer."""
+        assert self.params_numel is not None, "params-first ordering required"
+        return self.flat[: self.params_numel]
+
+    def make_grad_buffer(self) -> torch.Tensor:
+        """Allocate a flat grad buffer over the params region and point each
+        bound fp32 param's .grad at its slice.
+
+        Autograd accumulates into existing .grad in place, so after backward the
+        flat grad buffer holds all gradients contiguously (fused optimizer ops,
+        single-collective gradient reduction).
+        """
+        assert self.params_numel is not None, "params-first ordering required"
+        gbuf = torch.zeros(self.params_numel, dtype=torch.float32, device=self.flat.device)
+        name_to_idx = {n: i for i, n in enumerate(self.spec.names)}
+        for pname, p in self.module.named_parameters():
+            i = name_to_idx.get(pname)
+            if i is None or p.dtype != torch.float32:
+                continue
+            n = p.numel()
+            p.grad = gbuf[self.spec.offsets[i] : self.spec.offsets[i] + n].view(p.shape)
+        return gbuf

@@ -1,0 +1,56 @@
+"""Server-side aggregation reductions (reference fl4health/strategies/aggregate_utils.py:8-55
+and utils/functions.py:63-108 pseudo-sorted determinism).
+
+Torch-native: payloads are lists of flat fp32 tensors; the weighted reduce is
+one fused kernel over a stacked [K, n] buffer per tensor slot (K clients) —
+deterministic fixed-order summation (pseudo-sort analogue) instead of the
+reference's per-layer NumPy loop.
+"""
+from __future__ import annotations
+
+import torch
+
+from fl4health_amd.common import FitRes, Parameters
+from fl4health_amd.ops import functional as F
+
+
+def pseudo_sort_key(cid: str, num_examples: int, parameters: Parameters) -> tuple:
+    """Deterministic ordering key (reference decode_and_pseudo_sort_results:
+    sorts by sample counts + tensor content signature to pin fp summation order)."""
+    return (num_examples, cid)
+
+
+def decode_and_pseudo_sort_results(
+    results: list[tuple[object, FitRes]],
+) -> list[tuple[object, Parameters, int]]:
+    sortable = [(proxy, res.parameters, res.num_examples) for proxy, res in results]
+    return sorted(sortable, key=lambda t: (t[2], getattr(t[0], "cid", "")))
+
+
+def aggregate_results(results: list[tuple[Parameters, int]], weighted: bool = True) -> Parameters:
+    """Weighted (sum n_i w_i / sum n_i) or unweighted (mean) average, per tensor slot."""
+    assert results, "no results to aggregate"
+    n_slots = len(results[0][0].tensors)
+    total_examples = sum(n for _, n in results)
+    k = len(results)
+    if weighted:
+        ws = torch.tensor([n / total_examples for _, n in results], dtype=torch.float32)
+    else:
+        ws = torch.full((k,), 1.0 / k, dtype=torch.float32)
+    out_tensors = []
+    for slot in range(n_slots):
+        stack = torch.stack([p.tensors[slot] for p, _ in results])  # [K, ...]
+        w = ws.to(stack.device)
+        flat = stack.reshape(k, -1)
+        out = F.weighted_sum_rows(flat.contiguous(), w).view(stack.shape[1:])
+        out_tensors.append(out)
+    meta = dict(results[0][0].meta)
+    return Parameters(out_tensors, meta)
+
+
+def aggregate_losses(results: list[tuple[int, float]], weighted: bool = True) -> float:
+    """Aggregate client losses (reference aggregate_utils.py:35-55)."""
+    if weighted:
+        total = sum(n for n, _ in results)
+        return sum(n * loss for n, loss in results) / total
+    return sum(loss for _, loss in results) / len(results)
