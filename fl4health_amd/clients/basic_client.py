@@ -88,6 +88,10 @@ class BasicClient:
         self.reports_manager = ReportsManager(reporters)
         self.reports_manager.initialize(id=self.client_name, name="client")
 
+        # bf16 autocast for forward/backward (fp32 master weights + fp32
+        # flat aggregation); set autocast_dtype=torch.bfloat16 to enable
+        self.autocast_dtype: torch.dtype | None = None
+
         self.total_steps = 0
         self.total_epochs = 0
         self.current_server_round = 0
@@ -280,9 +284,15 @@ class BasicClient:
     def train_step(self, input: TorchInputType, target: TorchTargetType) -> tuple[TrainingLosses, TorchPredType]:
         """forward -> loss -> backward -> transform_gradients -> step (reference :578-603)."""
         self.set_optimizer_zero_grad()
-        preds, features = self.predict(input)
-        target = self.transform_target(target)
-        losses = self.compute_training_loss(preds, features, target)
+        if self.autocast_dtype is not None and self.device.type == "cuda":
+            with torch.autocast(device_type="cuda", dtype=self.autocast_dtype):
+                preds, features = self.predict(input)
+                target = self.transform_target(target)
+                losses = self.compute_training_loss(preds, features, target)
+        else:
+            preds, features = self.predict(input)
+            target = self.transform_target(target)
+            losses = self.compute_training_loss(preds, features, target)
         losses.backward["backward"].backward()
         self.transform_gradients(losses)
         self.step_optimizers()

@@ -81,7 +81,11 @@ class ScaffoldClient(BasicClient):
         assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
         model_params = self.parameter_exchanger.push_parameters(self.model, config=config)
         assert self.client_control_variates_updates is not None, "fit must run before get_parameters"
-        return self.parameter_exchanger.pack_parameters(model_params, self.client_control_variates_updates)
+        # variates live on the trainable params region; pad to full flat size so
+        # the packed wire payload [y || delta_c] has uniform slot shapes
+        full = torch.zeros_like(model_params.tensors[0])
+        full[: self.client_control_variates_updates.numel()] = self.client_control_variates_updates
+        return self.parameter_exchanger.pack_parameters(model_params, full)
 
 
 class DPScaffoldClient(ScaffoldClient):

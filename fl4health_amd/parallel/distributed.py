@@ -336,7 +336,19 @@ class DistributedRuntime:
             op = cmd["op"]
             if op == "shutdown":
                 return
-            if op == "fit":
+            if op == "sync":
+                self.sync()
+            elif op == "elapsed_gather":
+                self._all_gather_obj(self._elapsed)
+            elif op == "mark":
+                import time
+
+                self._mark = time.perf_counter()
+            elif op == "elapse":
+                import time
+
+                self._elapsed = time.perf_counter() - self._mark
+            elif op == "fit":
                 self._last_broadcast_params = self._bcast_parameters(None, src=0)
                 self._fit_body(cmd["cohort"], cmd["configs"], cmd["collective"], strategy)
             elif op == "evaluate":
@@ -352,3 +364,40 @@ class DistributedRuntime:
     # rank-0 wrapper so fit_clients sees the same broadcast params as workers
     def _record_broadcast(self, params: Parameters) -> None:
         self._last_broadcast_params = params
+
+    # ------------------------------------------------------------------
+    # benchmarking support: barrier+device-sync brackets and max-elapsed
+    # ------------------------------------------------------------------
+    _mark: float = 0.0
+    _elapsed: float = 0.0
+
+    def sync(self) -> None:
+        if self.comm_device.type == "cuda":
+            torch.cuda.synchronize(self.comm_device)
+        dist.barrier()
+        if self.comm_device.type == "cuda":
+            torch.cuda.synchronize(self.comm_device)
+
+    def bench_sync(self) -> None:
+        """Rank 0: bring every rank to a barrier + device sync."""
+        assert self.rank == 0
+        self._bcast_obj({"op": "sync"})
+        self.sync()
+
+    def bench_mark(self) -> None:
+        import time
+
+        assert self.rank == 0
+        self._bcast_obj({"op": "mark"})
+        self._mark = time.perf_counter()
+
+    def bench_elapsed_max(self) -> float:
+        """Rank 0: max over ranks of time since bench_mark."""
+        import time
+
+        assert self.rank == 0
+        self._bcast_obj({"op": "elapse"})
+        self._elapsed = time.perf_counter() - self._mark
+        self._bcast_obj({"op": "elapsed_gather"})
+        gathered = self._all_gather_obj(self._elapsed)
+        return max(float(g) for g in gathered)

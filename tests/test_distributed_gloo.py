@@ -1,0 +1,99 @@
+"""Multi-process distributed-path tests (gloo backend, world_size=2):
+validates the batched command protocol + collective aggregation against the
+in-process simulation result (role of the reference smoke tests' multi-process
+localhost-gRPC layer)."""
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+ROOT = Path(__file__).resolve().parent.parent
+
+WORKER = r"""
+import json, logging, torch
+logging.disable(logging.INFO)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.common import Parameters
+from fl4health_amd.clients.adaptive_drift_constraint_client import FedProxClient
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.optimizers import FlatProxSGD
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed
+from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+from tests.test_utils import TinyClient, TinyNet
+
+set_all_random_seeds(42)
+
+class Client(FedProxClient, TinyClient):
+    pass
+
+def strategy_factory():
+    init = Parameters([FlatParameterView(TinyNet()).flat.clone()])
+    return FedAvgWithAdaptiveConstraint(
+        initial_parameters=init, initial_loss_weight=0.1, adapt_loss_weight=True,
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 3})
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16}, strategy_factory())
+
+def client_factory(rank, world):
+    return Client(seed=rank, metrics=[Accuracy()], device="cpu")
+
+hist = run_distributed(server_factory, client_factory, num_rounds=2,
+                       strategy_factory=strategy_factory, backend="gloo")
+if hist is not None:
+    print("RESULT " + json.dumps({"losses": hist.losses_distributed}))
+"""
+
+
+def test_distributed_fedprox_collective_matches_simulation(tmp_path):
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29531",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    assert line, out.stdout[-3000:]
+    dist_losses = json.loads(line[0][7:])["losses"]
+    assert len(dist_losses) == 2
+
+    # in-process reference run with identical seeds/config
+    from fl4health_amd.utils.random import set_all_random_seeds
+    from fl4health_amd.client_managers.base import SimpleClientManager
+    from fl4health_amd.common import Parameters
+    from fl4health_amd.clients.adaptive_drift_constraint_client import FedProxClient
+    from fl4health_amd.metrics.metrics import Accuracy
+    from fl4health_amd.parameter_exchange.flat import FlatParameterView
+    from fl4health_amd.servers.base_server import FlServer
+    from fl4health_amd.simulation import run_simulation
+    from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+    from tests.test_utils import TinyClient, TinyNet
+
+    set_all_random_seeds(42)
+
+    class Client(FedProxClient, TinyClient):
+        pass
+
+    clients = [Client(seed=i, metrics=[Accuracy()], device="cpu") for i in range(2)]
+    init = Parameters([FlatParameterView(TinyNet()).flat.clone()])
+    strategy = FedAvgWithAdaptiveConstraint(
+        initial_parameters=init, initial_loss_weight=0.1, adapt_loss_weight=True,
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 3},
+    )
+    server = FlServer(SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16}, strategy)
+    hist = run_simulation(server, clients, num_rounds=2)
+
+    for (r1, l1), (r2, l2) in zip(dist_losses, hist.losses_distributed):
+        assert r1 == r2
+        assert abs(l1 - l2) < 5e-4, f"round {r1}: dist {l1} vs sim {l2}"

@@ -1,0 +1,63 @@
+import torch
+import torch.nn as nn
+
+from fl4health_amd.models.cnn import SmallCnn
+from fl4health_amd.models.resnet import ResNet18
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+
+
+def test_params_first_layout_and_roundtrip():
+    m = ResNet18()
+    view = FlatParameterView(m)
+    sd = m.state_dict()
+    assert view.params_numel == sum(p.numel() for p in m.parameters())
+    # first names are exactly the parameter names in order
+    pnames = [n for n, _ in m.named_parameters()]
+    assert view.spec.names[: len(pnames)] == pnames
+    # roundtrip
+    flat0 = view.clone_flat()
+    for p in m.parameters():
+        with torch.no_grad():
+            p.add_(1.0)
+    view.pull_into_flat()
+    assert not torch.allclose(view.flat, flat0)
+    view.load_flat(flat0)
+    view2 = FlatParameterView(m)
+    assert torch.allclose(view2.flat, flat0)
+
+
+def test_bound_views_share_storage():
+    m = SmallCnn()
+    view = FlatParameterView(m, bind=True)
+    # writing the flat buffer must be visible through module params
+    with torch.no_grad():
+        view.flat.zero_()
+    assert float(m.conv1.weight.abs().sum()) == 0.0
+    # forward+backward works on bound params
+    x = torch.randn(2, 3, 32, 32)
+    loss = m(x).sum()
+    loss.backward()
+    assert m.conv1.weight.grad is not None
+
+
+def test_grad_buffer_accumulates():
+    m = SmallCnn()
+    view = FlatParameterView(m, bind=True)
+    gbuf = view.make_grad_buffer()
+    x = torch.randn(2, 3, 32, 32)
+    m(x).sum().backward()
+    assert float(gbuf.abs().sum()) > 0
+    # grads landed in the flat buffer slices
+    assert m.conv1.weight.grad.data_ptr() == gbuf.data_ptr()
+
+
+def test_int_buffers_roundtrip():
+    m = nn.Sequential(nn.Linear(4, 4), nn.BatchNorm1d(4))
+    m.train()
+    m(torch.randn(8, 4))  # bumps num_batches_tracked
+    view = FlatParameterView(m)
+    nbt_before = int(m[1].num_batches_tracked)
+    flat = view.clone_flat()
+    m[1].num_batches_tracked.zero_()
+    view.load_flat(flat)
+    assert int(m[1].num_batches_tracked) == nbt_before

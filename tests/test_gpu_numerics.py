@@ -1,0 +1,168 @@
+"""GPU numerics: every HIP kernel vs the plain-PyTorch fp32 CPU oracle.
+
+All marked @pytest.mark.gpu (run on a real MI355X via gpurun). The op layer
+dispatches by device, so calling the same functional with CUDA tensors runs
+the hand-written CDNA4 kernel; CPU tensors run the torch reference.
+"""
+import pytest
+import torch
+
+from fl4health_amd.ops import functional as F
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+
+@requires_gpu
+def test_extension_loaded():
+    assert F.HAS_EXT, "HIP extension must be built and importable on the GPU box"
+
+
+@requires_gpu
+@pytest.mark.parametrize("n", [1, 255, 4096, 1 << 20, (1 << 22) + 3])
+def test_axpby_gpu(n):
+    x = torch.randn(n)
+    y = torch.randn(n)
+    yg = y.cuda()
+    F.axpby_(yg, x.cuda(), 1.7, -0.3)
+    F.axpby_(y, x, 1.7, -0.3)
+    assert torch.allclose(yg.cpu(), y, atol=1e-5)
+
+
+@requires_gpu
+def test_prox_sgd_gpu_matches_cpu():
+    n = 1 << 20
+    p = torch.randn(n)
+    g = torch.randn(n)
+    w0 = torch.randn(n)
+    m = torch.randn(n).abs()
+    pg, gg, w0g, mg = p.cuda(), g.cuda(), w0.cuda(), m.cuda()
+    for _ in range(3):
+        gn = torch.randn(n)
+        F.prox_sgd_step_(p, gn, w0, m, lr=0.05, mu=0.2, momentum=0.9, weight_decay=1e-3, nesterov=True)
+        F.prox_sgd_step_(pg, gn.cuda(), w0g, mg, lr=0.05, mu=0.2, momentum=0.9, weight_decay=1e-3, nesterov=True)
+    assert torch.allclose(pg.cpu(), p, atol=1e-4)
+    assert torch.allclose(mg.cpu(), m, atol=1e-4)
+
+
+@requires_gpu
+def test_scaffold_ops_gpu():
+    n = 1 << 18
+    p, g, c, ci = (torch.randn(n) for _ in range(4))
+    pg = p.cuda()
+    F.scaffold_sgd_step_(pg, g.cuda(), c.cuda(), ci.cuda(), lr=0.1)
+    F.scaffold_sgd_step_(p, g, c, ci, lr=0.1)
+    assert torch.allclose(pg.cpu(), p, atol=1e-5)
+
+    ci2 = torch.randn(n)
+    dci = torch.zeros(n)
+    ci2g, dcig = ci2.cuda(), torch.zeros(n, device="cuda")
+    x, y = torch.randn(n), torch.randn(n)
+    F.scaffold_variate_update_(ci2g, dcig, c.cuda(), x.cuda(), y.cuda(), inv_klr=4.0)
+    F.scaffold_variate_update_(ci2, dci, c, x, y, inv_klr=4.0)
+    assert torch.allclose(ci2g.cpu(), ci2, atol=1e-4)
+    assert torch.allclose(dcig.cpu(), dci, atol=1e-4)
+
+
+@requires_gpu
+@pytest.mark.parametrize("kind", ["fedavgm", "fedadam", "fedyogi", "fedadagrad", "flash"])
+def test_server_opt_gpu(kind):
+    n = 1 << 18
+    x = torch.randn(n)
+    xg = x.cuda()
+    m, v, d = torch.zeros(n), torch.zeros(n), torch.zeros(n)
+    mg, vg, dg = (t.cuda() for t in (m.clone(), v.clone(), d.clone()))
+    for _ in range(3):
+        delta = torch.randn(n)
+        F.server_opt_step_(x, delta, m, v, d, kind=kind, lr=0.1)
+        F.server_opt_step_(xg, delta.cuda(), mg, vg, dg, kind=kind, lr=0.1)
+    assert torch.allclose(xg.cpu(), x, atol=2e-4), f"{kind}: max diff {(xg.cpu()-x).abs().max()}"
+
+
+@requires_gpu
+def test_reductions_gpu_deterministic():
+    n = (1 << 22) + 17
+    x = torch.randn(n)
+    y = torch.randn(n)
+    sq_cpu = float(F.sq_norm(x))
+    sq_gpu1 = float(F.sq_norm(x.cuda()))
+    sq_gpu2 = float(F.sq_norm(x.cuda()))
+    assert sq_gpu1 == sq_gpu2, "GPU reduction must be bitwise deterministic"
+    assert abs(sq_cpu - sq_gpu1) / abs(sq_cpu) < 1e-6
+    assert abs(float(F.dot(x.cuda(), y.cuda())) - float(F.dot(x, y))) < 1e-2
+
+
+@requires_gpu
+def test_clip_delta_gpu():
+    n = 1 << 16
+    w = torch.randn(n)
+    w0 = torch.randn(n)
+    out_c, bit_c = F.clip_delta(w, w0, clip_bound=5.0)
+    out_g, bit_g = F.clip_delta(w.cuda(), w0.cuda(), clip_bound=5.0)
+    assert torch.allclose(out_g.cpu(), out_c, atol=1e-5)
+    assert float(bit_g[0]) == float(bit_c[0])
+
+
+@requires_gpu
+def test_gaussian_noise_gpu_stats_and_replay():
+    n = 1 << 20
+    x1 = torch.zeros(n, device="cuda")
+    x2 = torch.zeros(n, device="cuda")
+    F.gaussian_noise_(x1, sigma=2.0, seed=123, offset=0)
+    F.gaussian_noise_(x2, sigma=2.0, seed=123, offset=0)
+    assert torch.equal(x1, x2), "counter-based noise must replay exactly"
+    assert abs(float(x1.mean())) < 0.02
+    assert abs(float(x1.std()) - 2.0) < 0.02
+    x3 = torch.zeros(n, device="cuda")
+    F.gaussian_noise_(x3, sigma=2.0, seed=124, offset=0)
+    assert not torch.equal(x1, x3)
+
+
+@requires_gpu
+def test_bernoulli_mask_gpu_stats():
+    n = 1 << 20
+    scores = torch.full((n,), 0.5, device="cuda")
+    mask, weff = F.bernoulli_mask(scores, torch.ones(n, device="cuda"), seed=9)
+    p = torch.sigmoid(torch.tensor(0.5))
+    assert abs(float(mask.mean()) - float(p)) < 5e-3
+    assert torch.equal(weff, mask)
+    mask2, _ = F.bernoulli_mask(scores, None, seed=9)
+    assert torch.equal(mask, mask2)
+
+
+@requires_gpu
+def test_per_sample_clip_gpu():
+    B, D = 32, 100000
+    g = torch.randn(B, D)
+    norms_c = torch.zeros(B)
+    F.per_sample_sqnorm_(g, norms_c)
+    norms_g = torch.zeros(B, device="cuda")
+    F.per_sample_sqnorm_(g.cuda(), norms_g)
+    assert torch.allclose(norms_g.cpu(), norms_c, rtol=1e-4)
+    out_c = torch.zeros(D)
+    out_g = torch.zeros(D, device="cuda")
+    F.clip_rowsum_(g, norms_c, out_c, clip_bound=1.0)
+    F.clip_rowsum_(g.cuda(), norms_g, out_g, clip_bound=1.0)
+    assert torch.allclose(out_g.cpu(), out_c, atol=1e-3)
+
+
+@requires_gpu
+def test_confusion_counts_gpu():
+    n = 1 << 20
+    preds = torch.randint(0, 7, (n,))
+    tgts = torch.randint(0, 7, (n,))
+    out_c = torch.zeros(7, 4, dtype=torch.int64)
+    F.confusion_counts_(preds, tgts, out_c)
+    out_g = torch.zeros(7, 4, dtype=torch.int64, device="cuda")
+    F.confusion_counts_(preds.cuda(), tgts.cuda(), out_g)
+    assert torch.equal(out_g.cpu(), out_c)
+
+
+@requires_gpu
+def test_weighted_sum_rows_gpu():
+    stack = torch.randn(8, 1 << 20)
+    w = torch.rand(8)
+    out_c = F.weighted_sum_rows(stack, w)
+    out_g = F.weighted_sum_rows(stack.cuda(), w.cuda())
+    assert torch.allclose(out_g.cpu(), out_c, atol=1e-4)

@@ -1,0 +1,171 @@
+"""CPU-oracle tests for the op layer semantics (the same functions dispatch to
+HIP kernels on GPU; tests/test_gpu_numerics.py compares the two)."""
+import numpy as np
+import pytest
+import torch
+
+from fl4health_amd.ops import functional as F
+
+
+def test_axpby():
+    y = torch.randn(100)
+    x = torch.randn(100)
+    ref = 2.0 * x + 0.5 * y
+    F.axpby_(y, x, 2.0, 0.5)
+    assert torch.allclose(y, ref, atol=1e-6)
+
+
+def test_prox_sgd_matches_manual():
+    p = torch.randn(50)
+    g = torch.randn(50)
+    w0 = torch.randn(50)
+    m = torch.zeros(50)
+    p0 = p.clone()
+    F.prox_sgd_step_(p, g, w0, m, lr=0.1, mu=0.3, momentum=0.9, weight_decay=0.01)
+    geff = g + 0.01 * p0 + 0.3 * (p0 - w0)
+    mref = geff.clone()
+    pref = p0 - 0.1 * mref
+    assert torch.allclose(p, pref, atol=1e-6)
+    assert torch.allclose(m, mref, atol=1e-6)
+    # second step exercises momentum accumulation
+    g2 = torch.randn(50)
+    p1 = p.clone()
+    F.prox_sgd_step_(p, g2, w0, m, lr=0.1, mu=0.3, momentum=0.9, weight_decay=0.01)
+    geff2 = g2 + 0.01 * p1 + 0.3 * (p1 - w0)
+    mref2 = 0.9 * mref + geff2
+    assert torch.allclose(p, p1 - 0.1 * mref2, atol=1e-6)
+
+
+def test_prox_sgd_matches_torch_sgd_when_mu_zero():
+    torch.manual_seed(0)
+    w = torch.randn(64, requires_grad=True)
+    opt = torch.optim.SGD([w], lr=0.05, momentum=0.9, weight_decay=1e-2)
+    p = w.detach().clone()
+    m = torch.zeros_like(p)
+    for step in range(3):
+        g = torch.randn(64)
+        w.grad = g.clone()
+        opt.step()
+        F.prox_sgd_step_(p, g, None, m, lr=0.05, mu=0.0, momentum=0.9, weight_decay=1e-2)
+        assert torch.allclose(p, w.detach(), atol=1e-5), f"step {step}"
+
+
+def test_scaffold_sgd_step():
+    p = torch.randn(30)
+    g = torch.randn(30)
+    c = torch.randn(30)
+    ci = torch.randn(30)
+    p0 = p.clone()
+    F.scaffold_sgd_step_(p, g, c, ci, lr=0.1)
+    assert torch.allclose(p, p0 - 0.1 * (g + c - ci), atol=1e-6)
+
+
+def test_scaffold_variate_update():
+    ci = torch.randn(20)
+    dci = torch.zeros(20)
+    c = torch.randn(20)
+    x = torch.randn(20)
+    y = torch.randn(20)
+    ci0 = ci.clone()
+    F.scaffold_variate_update_(ci, dci, c, x, y, inv_klr=1.0 / (5 * 0.05))
+    ci_ref = ci0 - c + (x - y) / (5 * 0.05)
+    assert torch.allclose(ci, ci_ref, atol=1e-5)
+    assert torch.allclose(dci, ci_ref - ci0, atol=1e-5)
+
+
+def test_reductions():
+    x = torch.randn(1000)
+    y = torch.randn(1000)
+    assert abs(float(F.sq_norm(x)) - float((x**2).sum())) < 1e-3
+    assert abs(float(F.sq_diff(x, y)) - float(((x - y) ** 2).sum())) < 1e-3
+    assert abs(float(F.dot(x, y)) - float((x * y).sum())) < 1e-3
+
+
+def test_clip_delta():
+    w0 = torch.zeros(100)
+    w = torch.ones(100) * 2.0  # norm = 20
+    out, bit = F.clip_delta(w, w0, clip_bound=10.0)
+    assert abs(float(out.norm()) - 10.0) < 1e-4
+    assert float(bit[0]) == 0.0  # NOT within bound
+    out2, bit2 = F.clip_delta(w, w0, clip_bound=100.0)
+    assert torch.allclose(out2, w - w0)
+    assert float(bit2[0]) == 1.0
+
+
+def test_server_opt_flash_matches_numpy_reference():
+    """Flash moment math vs a direct NumPy transcription of the reference
+    update rule (fl4health/strategies/flash.py:125-170)."""
+    n = 64
+    rng = np.random.default_rng(0)
+    x = rng.normal(size=n).astype(np.float32)
+    m = np.zeros(n, dtype=np.float32)
+    v = np.zeros(n, dtype=np.float32)
+    d = np.zeros(n, dtype=np.float32)
+    xt = torch.tensor(x.copy())
+    mt, vt, dt = torch.zeros(n), torch.zeros(n), torch.zeros(n)
+    b1, b2, eta, tau = 0.9, 0.99, 0.1, 1e-9
+    for _ in range(3):
+        delta = rng.normal(size=n).astype(np.float32)
+        d2 = delta * delta
+        m = b1 * m + (1 - b1) * delta
+        v_prev = v.copy()
+        v = b2 * v + (1 - b2) * d2
+        diff = d2 - v
+        denom = np.abs(diff) + np.abs(v_prev)
+        b3 = np.where(denom > 0, np.abs(v_prev) / denom, 0.0)
+        d = b3 * d + (1 - b3) * diff
+        x = x + eta * m / (np.sqrt(v) - d + tau)
+        F.server_opt_step_(xt, torch.tensor(delta), mt, vt, dt, kind="flash", beta1=b1, beta2=b2, lr=eta, tau=tau)
+    assert torch.allclose(xt, torch.tensor(x), atol=1e-4)
+
+
+@pytest.mark.parametrize("kind", ["fedavgm", "fedadam", "fedyogi", "fedadagrad"])
+def test_server_opt_kinds_run(kind):
+    x = torch.randn(32)
+    delta = torch.randn(32)
+    m, v, d = torch.zeros(32), torch.zeros(32), torch.zeros(32)
+    x0 = x.clone()
+    F.server_opt_step_(x, delta, m, v, d, kind=kind, lr=0.1)
+    assert not torch.allclose(x, x0)
+    assert torch.isfinite(x).all()
+
+
+def test_confusion_counts():
+    preds = torch.tensor([0, 1, 2, 1, 0, 2, 2])
+    tgts = torch.tensor([0, 1, 1, 1, 2, 2, 0])
+    out = torch.zeros(3, 4, dtype=torch.int64)
+    F.confusion_counts_(preds, tgts, out)
+    # class 0: tp=1 (idx0), fp=1 (idx4 pred0 tgt2), fn=1 (idx6 tgt0 pred2)
+    assert out[0].tolist() == [1, 1, 1, 4]
+    # class 1: tp=2, fp=0, fn=1 (idx2 tgt1 pred2)
+    assert out[1].tolist() == [2, 0, 1, 4]
+    # class 2: tp=1, fp=2, fn=1
+    assert out[2].tolist() == [1, 2, 1, 3]
+
+
+def test_bernoulli_mask_stats():
+    scores = torch.full((20000,), 2.0)  # sigmoid(2) ~ 0.881
+    mask, weff = F.bernoulli_mask(scores, torch.ones(20000), seed=7)
+    rate = float(mask.mean())
+    assert abs(rate - 0.8808) < 0.02
+    assert torch.equal(weff, mask)
+
+
+def test_per_sample_clip_pipeline():
+    B, D = 4, 50
+    g = torch.randn(B, D)
+    norms = torch.zeros(B)
+    F.per_sample_sqnorm_(g, norms)
+    assert torch.allclose(norms, g.pow(2).sum(dim=1), atol=1e-4)
+    out = torch.zeros(D)
+    F.clip_rowsum_(g, norms, out, clip_bound=1.0)
+    coef = torch.clamp(1.0 / (norms.sqrt() + 1e-6), max=1.0)
+    ref = (coef.unsqueeze(1) * g).sum(dim=0)
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_weighted_sum_rows():
+    stack = torch.randn(3, 40)
+    w = torch.tensor([0.2, 0.3, 0.5])
+    out = F.weighted_sum_rows(stack, w)
+    assert torch.allclose(out, (w.unsqueeze(1) * stack).sum(0), atol=1e-6)
