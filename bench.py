@@ -46,6 +46,7 @@ class BenchFedProxClient(FedProxClient):
         self.args = args
         if self.device.type == "cuda":
             self.autocast_dtype = torch.bfloat16
+            self.use_cuda_graph = not args.no_graph
 
     def get_model(self, config):
         return ResNet18(num_classes=10)
@@ -99,6 +100,7 @@ def main() -> None:
     parser.add_argument("--local_steps", type=int, default=5)
     parser.add_argument("--batch_size", type=int, default=128)
     parser.add_argument("--shard_size", type=int, default=8192)
+    parser.add_argument("--no_graph", action="store_true", help="disable hipGraph train-step capture")
     args = parser.parse_args()
 
     set_all_random_seeds(42)

@@ -34,8 +34,10 @@ class SimpleMetric(Metric, ABC):
         self.accumulated_targets: list[torch.Tensor] = []
 
     def update(self, input: torch.Tensor, target: torch.Tensor) -> None:
-        self.accumulated_inputs.append(input.detach())
-        self.accumulated_targets.append(target.detach())
+        # clone: under hipGraph replay the incoming preds are static graph
+        # output buffers whose contents change on every replay
+        self.accumulated_inputs.append(input.detach().clone())
+        self.accumulated_targets.append(target.detach().clone())
 
     def compute(self, name: str | None = None) -> dict[str, Scalar]:
         if not self.accumulated_inputs:

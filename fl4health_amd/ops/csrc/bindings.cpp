@@ -5,8 +5,8 @@
 
 extern "C" {
 void launch_axpby(float*, const float*, float, float, int64_t, hipStream_t);
-void launch_prox_sgd(float*, const float*, const float*, float*, float, float, float, float, int,
-                     int64_t, hipStream_t);
+void launch_prox_sgd(float*, const float*, const float*, float*, float, float, const float*,
+                     float, float, int, int64_t, hipStream_t);
 void launch_scaffold_sgd(float*, const float*, const float*, const float*, float, float, int64_t,
                          hipStream_t);
 void launch_scaffold_variate(float*, float*, const float*, const float*, const float*, float,
@@ -44,15 +44,18 @@ void axpby_(torch::Tensor y, torch::Tensor x, double a, double b) {
 }
 
 void prox_sgd_step_(torch::Tensor p, torch::Tensor g, c10::optional<torch::Tensor> w0,
-                    c10::optional<torch::Tensor> mbuf, double lr, double mu, double momentum,
-                    double weight_decay, bool nesterov) {
+                    c10::optional<torch::Tensor> mbuf, double lr, double mu,
+                    c10::optional<torch::Tensor> mu_dev, double momentum, double weight_decay,
+                    bool nesterov) {
   check_f32(p, "p");
   check_f32(g, "g");
   const float* w0p = nullptr;
   float* mp = nullptr;
+  const float* mudp = nullptr;
   if (w0.has_value()) { check_f32(*w0, "w0"); w0p = w0->data_ptr<float>(); }
   if (mbuf.has_value()) { check_f32(*mbuf, "mbuf"); mp = mbuf->data_ptr<float>(); }
-  launch_prox_sgd(p.data_ptr<float>(), g.data_ptr<float>(), w0p, mp, (float)lr, (float)mu,
+  if (mu_dev.has_value()) { check_f32(*mu_dev, "mu_dev"); mudp = mu_dev->data_ptr<float>(); }
+  launch_prox_sgd(p.data_ptr<float>(), g.data_ptr<float>(), w0p, mp, (float)lr, (float)mu, mudp,
                   (float)momentum, (float)weight_decay, nesterov ? 1 : 0, p.numel(), stream());
 }
 

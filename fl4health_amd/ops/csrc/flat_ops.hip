@@ -68,8 +68,11 @@ extern "C" void launch_axpby(float* y, const float* x, float a, float b, int64_t
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(BLOCK) void prox_sgd_kernel(
     float* __restrict__ p, const float* __restrict__ g, const float* __restrict__ w0,
-    float* __restrict__ mbuf, float lr, float mu, float momentum, float weight_decay,
-    int nesterov, int64_t n) {
+    float* __restrict__ mbuf, float lr, float mu, const float* __restrict__ mu_dev,
+    float momentum, float weight_decay, int nesterov, int64_t n) {
+  // mu may live in device memory (mu_dev) so a hipGraph-captured step sees
+  // per-round server-adapted mu without re-capture
+  if (mu_dev != nullptr) mu = mu_dev[0];
   GSL(i, n, STRIDE) {
     float gi = g[i];
     float pi = p[i];
@@ -86,9 +89,9 @@ __global__ __launch_bounds__(BLOCK) void prox_sgd_kernel(
 }
 
 extern "C" void launch_prox_sgd(float* p, const float* g, const float* w0, float* mbuf,
-                                float lr, float mu, float momentum, float weight_decay,
-                                int nesterov, int64_t n, hipStream_t s) {
-  prox_sgd_kernel<<<grid_1d(n), BLOCK, 0, s>>>(p, g, w0, mbuf, lr, mu, momentum,
+                                float lr, float mu, const float* mu_dev, float momentum,
+                                float weight_decay, int nesterov, int64_t n, hipStream_t s) {
+  prox_sgd_kernel<<<grid_1d(n), BLOCK, 0, s>>>(p, g, w0, mbuf, lr, mu, mu_dev, momentum,
                                                weight_decay, nesterov, n);
 }
 

@@ -56,21 +56,29 @@ def prox_sgd_step_(
     mbuf: torch.Tensor | None = None,
     *,
     lr: float,
-    mu: float = 0.0,
+    mu: float | torch.Tensor = 0.0,
     momentum: float = 0.0,
     weight_decay: float = 0.0,
     nesterov: bool = False,
 ) -> None:
-    """Fused (proximal) SGD step: g' = g + wd*p + mu*(p - w0); momentum; p -= lr*u."""
+    """Fused (proximal) SGD step: g' = g + wd*p + mu*(p - w0); momentum; p -= lr*u.
+
+    ``mu`` may be a device 0/1-dim fp32 tensor: the kernel reads it from device
+    memory, so a hipGraph-captured step tracks per-round mu adaptation.
+    """
     if p.is_cuda:
         _require_ext("prox_sgd_step_")
-        _C.prox_sgd_step_(p, g, w0, mbuf, lr, mu, momentum, weight_decay, nesterov)
+        if isinstance(mu, torch.Tensor):
+            _C.prox_sgd_step_(p, g, w0, mbuf, lr, 0.0, mu.reshape(1), momentum, weight_decay, nesterov)
+        else:
+            _C.prox_sgd_step_(p, g, w0, mbuf, lr, mu, None, momentum, weight_decay, nesterov)
         return
+    mu_f = float(mu.item()) if isinstance(mu, torch.Tensor) else mu
     geff = g.clone()
     if weight_decay != 0.0:
         geff.add_(p, alpha=weight_decay)
-    if w0 is not None and mu != 0.0:
-        geff.add_(p - w0, alpha=mu)
+    if w0 is not None and mu_f != 0.0:
+        geff.add_(p - w0, alpha=mu_f)
     u = geff
     if mbuf is not None:
         mbuf.mul_(momentum).add_(geff)

@@ -65,29 +65,50 @@ class FlatProxSGD(FlatOptimizerBase):
         self.momentum = momentum
         self.weight_decay = weight_decay
         self.nesterov = nesterov
-        self.mu = mu
         self.mbuf = torch.zeros_like(self.gbuf) if momentum != 0.0 else None
-        self.w0: torch.Tensor | None = None  # proximal anchor (params region)
+        self._on_cuda = self.view.flat.is_cuda
+        # persistent anchor buffer + device-resident mu: both pointers stay
+        # FIXED so a hipGraph capture of the train step remains valid across
+        # rounds (set_anchor/set_penalty_weight update contents, not pointers)
+        self.w0: torch.Tensor | None = None
+        if self._on_cuda:
+            self.mu: "torch.Tensor | float" = torch.full((1,), float(mu), dtype=torch.float32, device=self.view.flat.device)
+        else:
+            self.mu = float(mu)
 
     def set_anchor(self, w0: torch.Tensor | None) -> None:
         """Set the proximal anchor weights (round-start global params)."""
-        self.w0 = w0
+        if w0 is None:
+            self.w0 = None
+            return
+        if self.w0 is None:
+            self.w0 = w0.detach().clone().to(self.view.flat.device)
+        else:
+            self.w0.copy_(w0)
 
     def set_penalty_weight(self, mu: float) -> None:
-        self.mu = mu
+        if isinstance(self.mu, torch.Tensor):
+            self.mu.fill_(float(mu))
+        else:
+            self.mu = float(mu)
+
+    def _mu_value(self) -> float:
+        return float(self.mu.item()) if isinstance(self.mu, torch.Tensor) else self.mu
 
     def drift_loss(self) -> torch.Tensor:
         """mu/2 * ||w - w0||^2 (device scalar; reference weight_drift_loss.py)."""
-        if self.w0 is None or self.mu == 0.0:
+        if self.w0 is None:
             return torch.zeros((), device=self.view.flat.device)
-        return 0.5 * self.mu * F.sq_diff(self.view.params_region, self.w0).to(torch.float32)
+        sq = F.sq_diff(self.view.params_region, self.w0).to(torch.float32)
+        mu = self.mu if isinstance(self.mu, torch.Tensor) else torch.tensor(self.mu)
+        return 0.5 * mu.reshape(()).to(sq.device) * sq
 
     @torch.no_grad()
     def step(self, closure=None) -> None:  # noqa: ARG002
         F.prox_sgd_step_(
             self.view.params_region,
             self.gbuf,
-            self.w0 if self.mu != 0.0 else None,
+            self.w0,
             self.mbuf,
             lr=self.lr,
             mu=self.mu,
@@ -97,12 +118,13 @@ class FlatProxSGD(FlatOptimizerBase):
         )
 
     def _extra_state(self) -> dict[str, Any]:
-        return {"mbuf": self.mbuf, "mu": self.mu, "lr": self.lr}
+        return {"mbuf": self.mbuf, "mu": self._mu_value(), "lr": self.lr}
 
     def _load_extra_state(self, extra: dict[str, Any]) -> None:
         if extra.get("mbuf") is not None and self.mbuf is not None:
             self.mbuf.copy_(extra["mbuf"])
-        self.mu = extra.get("mu", self.mu)
+        if "mu" in extra:
+            self.set_penalty_weight(extra["mu"])
         self.lr = extra.get("lr", self.lr)
 
 
