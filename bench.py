@@ -52,21 +52,24 @@ class BenchFedProxClient(FedProxClient):
         return ResNet18(num_classes=10)
 
     def get_data_loaders(self, config):
-        from torch.utils.data import DataLoader
+        from fl4health_amd.datasets.loaders import DeviceTensorLoader
+        from fl4health_amd.datasets.partitioners import DirichletLabelPartitioner
 
         n_total = self.args.shard_size * self.world
         ds = synthetic_classification_dataset(n_total, (3, 32, 32), 10, seed=1234)
-        labels = ds.tensors[1]
-        shards = DirichletLabelPartitioner(self.world, beta=0.5, min_size=8, seed=1234).partition_dataset(ds, labels)
-        shard = shards[self.rank]
-        train_loader = DataLoader(
-            shard,
+        x, labels = ds.tensors
+        idx = DirichletLabelPartitioner(self.world, beta=0.5, min_size=8, seed=1234).partition_indices(labels)[self.rank]
+        idx_t = torch.as_tensor(idx)
+        # shard stays resident in HBM; batches are device gathers (no host loop)
+        train_loader = DeviceTensorLoader(
+            x.index_select(0, idx_t),
+            labels.index_select(0, idx_t),
             batch_size=self.args.batch_size,
+            device=self.device,
             shuffle=True,
-            generator=torch.Generator().manual_seed(42 + self.rank),
             drop_last=True,
-            num_workers=0,
-            pin_memory=self.device.type == "cuda",
+            seed=42 + self.rank,
+            channels_last=self.device.type == "cuda",
         )
         return train_loader, None
 
@@ -105,6 +108,7 @@ def main() -> None:
 
     set_all_random_seeds(42)
     logging.basicConfig(level=logging.WARNING)
+    torch.backends.cudnn.benchmark = True  # MIOpen algo find during warmup
 
     has_gpu = torch.cuda.is_available()
     runtime = DistributedRuntime(backend="nccl" if has_gpu else "gloo")
