@@ -1,0 +1,48 @@
+"""Clipping client for client-level DP (reference fl4health/clients/clipping_client.py:22-188).
+
+Computes the round weight-delta vs round-start, flat-clips its l2 norm to the
+server-broadcast bound C, and packs the clipping bit. The norm + scale is the
+fused clip_delta kernel pair over the flat buffer (K5).
+"""
+from __future__ import annotations
+
+import torch
+
+from fl4health_amd.clients.basic_client import BasicClient
+from fl4health_amd.common import Config, Parameters
+from fl4health_amd.ops import functional as F
+from fl4health_amd.parameter_exchange.exchangers import FullParameterExchangerWithPacking
+from fl4health_amd.parameter_exchange.packers import ParameterPackerWithClippingBit
+
+
+class NumpyClippingClient(BasicClient):
+    """Name retained from the reference API; payloads are flat torch tensors here."""
+
+    def __init__(self, *args, adaptive_clipping: bool = False, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        self.adaptive_clipping = adaptive_clipping
+        self.clipping_bound: float | None = None
+        self.initial_flat: torch.Tensor | None = None
+
+    def get_parameter_exchanger(self, config: Config) -> FullParameterExchangerWithPacking:
+        return FullParameterExchangerWithPacking(ParameterPackerWithClippingBit())
+
+    def set_parameters(self, parameters: Parameters, config: Config, fitting_round: bool) -> None:
+        assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
+        model_params, clipping_bound = self.parameter_exchanger.unpack_parameters(parameters)
+        self.clipping_bound = float(clipping_bound)
+        self.parameter_exchanger.pull_parameters(model_params, self.model, config)
+        if fitting_round:
+            self.initial_flat = self.flat_view.clone_flat()
+
+    def compute_weight_update_and_clip(self) -> tuple[torch.Tensor, float]:
+        assert self.initial_flat is not None and self.clipping_bound is not None
+        self.flat_view.pull_into_flat()
+        clipped_delta, bit = F.clip_delta(self.flat_view.flat, self.initial_flat, self.clipping_bound)
+        return clipped_delta, float(bit[0].item()) if bit is not None else 1.0
+
+    def get_parameters(self, config: Config) -> Parameters:
+        assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
+        clipped_delta, bit = self.compute_weight_update_and_clip()
+        # client-level DP sends clipped weight DELTAS, not weights
+        return self.parameter_exchanger.pack_parameters(Parameters([clipped_delta]), bit)

@@ -39,7 +39,9 @@ class FullParameterExchanger(ParameterExchanger):
         self._view: FlatParameterView | None = None
 
     def view_for(self, model: nn.Module) -> FlatParameterView:
-        if self._view is None or self._view.module is not model:
+        # an explicitly attached view (client wiring, possibly over a wrapped
+        # module such as GradSampleModule's inner model) takes precedence
+        if self._view is None:
             self._view = FlatParameterView(model)
         return self._view
 

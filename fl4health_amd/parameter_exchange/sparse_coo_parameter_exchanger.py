@@ -1,0 +1,55 @@
+"""SparseCooParameterExchanger (reference fl4health/parameter_exchange/
+sparse_coo_parameter_exchanger.py:18-162): score-function-driven arbitrary
+parameter-subset exchange in sparse COO form (K11)."""
+from __future__ import annotations
+
+from typing import Callable
+
+import torch
+import torch.nn as nn
+
+from fl4health_amd.common import Config, Parameters
+from fl4health_amd.parameter_exchange.exchangers import ParameterExchanger
+from fl4health_amd.parameter_exchange.packers import SparseCooParameterPacker
+
+ScoreGenFunction = Callable[[nn.Module, nn.Module | None], dict[str, torch.Tensor]]
+
+
+class SparseCooParameterExchanger(ParameterExchanger):
+    def __init__(self, sparsity_level: float, score_gen_function: ScoreGenFunction) -> None:
+        assert 0.0 < sparsity_level <= 1.0
+        self.sparsity_level = sparsity_level
+        self.score_gen_function = score_gen_function
+        self.packer = SparseCooParameterPacker()
+
+    def select_parameters(self, model: nn.Module, initial_model: nn.Module | None = None):
+        scores = self.score_gen_function(model, initial_model)
+        all_scores = torch.cat([s.reshape(-1) for s in scores.values()])
+        k = max(int(all_scores.numel() * self.sparsity_level), 1)
+        threshold = torch.kthvalue(all_scores.float().cpu(), all_scores.numel() - k + 1).values
+        sd = model.state_dict()
+        values, indices, shapes, names = [], [], [], []
+        for name, score in scores.items():
+            mask = score >= threshold.to(score.device)
+            nz = mask.nonzero().t()
+            values.append(sd[name][mask].detach().reshape(-1).float())
+            indices.append(nz)
+            shapes.append(list(score.shape))
+            names.append(name)
+        return values, indices, shapes, names
+
+    def push_parameters(self, model: nn.Module, initial_model: nn.Module | None = None, config: Config | None = None) -> Parameters:
+        values, indices, shapes, names = self.select_parameters(model, initial_model)
+        return self.packer.pack_parameters(
+            Parameters([]), {"values": values, "indices": indices, "shapes": shapes, "names": names}
+        )
+
+    def pull_parameters(self, parameters: Parameters, model: nn.Module, config: Config | None = None) -> None:
+        _, info = self.packer.unpack_parameters(parameters)
+        sd = model.state_dict()
+        with torch.no_grad():
+            for name, vals, idx in zip(info["names"], info["values"], info["indices"]):
+                t = sd[name]
+                if idx.numel() == 0:
+                    continue
+                t[tuple(idx.long())] = vals.to(t.device, t.dtype)

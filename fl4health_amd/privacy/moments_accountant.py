@@ -1,0 +1,104 @@
+"""Moments (RDP) accountant for the subsampled Gaussian mechanism.
+
+Capability of reference fl4health/privacy/moments_accountant.py:30-160, which
+wraps Google `dp-accounting` (unavailable offline): from-scratch RDP of the
+Poisson-subsampled Gaussian at integer orders (Mironov et al. 2019, Thm. 11 /
+the standard binomial-expansion upper bound), composed across steps, with the
+classic RDP->(eps, delta) conversion.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+DEFAULT_ORDERS = list(range(2, 65)) + [80, 96, 128, 192, 256, 512]
+
+
+def _log_add(a: float, b: float) -> float:
+    if a == -math.inf:
+        return b
+    if b == -math.inf:
+        return a
+    m = max(a, b)
+    return m + math.log1p(math.exp(min(a, b) - m))
+
+
+def _log_comb(n: int, k: int) -> float:
+    return math.lgamma(n + 1) - math.lgamma(k + 1) - math.lgamma(n - k + 1)
+
+
+def rdp_subsampled_gaussian(q: float, sigma: float, alpha: int) -> float:
+    """RDP epsilon at integer order alpha for sampling rate q, noise sigma."""
+    if q == 0.0:
+        return 0.0
+    if sigma == 0.0:
+        return math.inf
+    if q == 1.0:
+        return alpha / (2 * sigma**2)
+    # log E[( (1-q) + q e^{(k... )})] binomial expansion:
+    # A(alpha) = sum_{k=0}^{alpha} C(alpha,k) (1-q)^{alpha-k} q^k exp(k(k-1)/(2 sigma^2))
+    log_a = -math.inf
+    for k in range(alpha + 1):
+        term = (
+            _log_comb(alpha, k)
+            + (alpha - k) * math.log1p(-q)
+            + (k * math.log(q) if q > 0 else -math.inf)
+            + k * (k - 1) / (2 * sigma**2)
+        )
+        log_a = _log_add(log_a, term)
+    return log_a / (alpha - 1)
+
+
+@dataclass
+class PoissonSampling:
+    sampling_probability: float
+
+
+@dataclass
+class FixedSamplingWithoutReplacement:
+    population_size: int
+    sample_size: int
+
+    @property
+    def sampling_probability(self) -> float:
+        return self.sample_size / self.population_size
+
+
+class MomentsAccountant:
+    def __init__(self, moment_orders: list[int] | None = None) -> None:
+        self.orders = moment_orders or DEFAULT_ORDERS
+
+    def _total_rdp(self, sampling_rates: list[float], sigmas: list[float], steps: list[int]) -> list[float]:
+        totals = [0.0] * len(self.orders)
+        for q, sigma, n in zip(sampling_rates, sigmas, steps):
+            for i, alpha in enumerate(self.orders):
+                totals[i] += n * rdp_subsampled_gaussian(q, sigma, alpha)
+        return totals
+
+    def get_epsilon(
+        self, sampling_rates: list[float] | float, noise_multipliers: list[float] | float,
+        steps: list[int] | int, delta: float,
+    ) -> float:
+        qs = sampling_rates if isinstance(sampling_rates, list) else [sampling_rates]
+        sigmas = noise_multipliers if isinstance(noise_multipliers, list) else [noise_multipliers] * len(qs)
+        ns = steps if isinstance(steps, list) else [steps] * len(qs)
+        rdp = self._total_rdp(qs, sigmas, ns)
+        eps = math.inf
+        for alpha, r in zip(self.orders, rdp):
+            if math.isfinite(r):
+                eps = min(eps, r + math.log(1.0 / delta) / (alpha - 1))
+        return eps
+
+    def get_delta(
+        self, sampling_rates: list[float] | float, noise_multipliers: list[float] | float,
+        steps: list[int] | int, epsilon: float,
+    ) -> float:
+        qs = sampling_rates if isinstance(sampling_rates, list) else [sampling_rates]
+        sigmas = noise_multipliers if isinstance(noise_multipliers, list) else [noise_multipliers] * len(qs)
+        ns = steps if isinstance(steps, list) else [steps] * len(qs)
+        rdp = self._total_rdp(qs, sigmas, ns)
+        delta = 1.0
+        for alpha, r in zip(self.orders, rdp):
+            if math.isfinite(r):
+                delta = min(delta, math.exp((alpha - 1) * (r - epsilon)))
+        return delta

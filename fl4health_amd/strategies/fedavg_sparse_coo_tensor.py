@@ -1,0 +1,65 @@
+"""FedAvgSparseCooTensor (reference fl4health/strategies/fedavg_sparse_coo_tensor.py:18-316):
+per-tensor aggregation of sparse COO-packed params; only the clients that
+sent a coordinate contribute to its average (dense scatter-accumulate on
+device, K11)."""
+from __future__ import annotations
+
+import torch
+
+from fl4health_amd.client_managers.base import ClientProxy
+from fl4health_amd.common import FitRes, Parameters, Scalar
+from fl4health_amd.parameter_exchange.packers import SparseCooParameterPacker
+from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+
+
+class FedAvgSparseCooTensor(BasicFedAvg):
+    def __init__(self, **kwargs) -> None:
+        kwargs.setdefault("weighted_aggregation", False)
+        super().__init__(**kwargs)
+        self.parameter_packer = SparseCooParameterPacker()
+
+    def aggregate_fit(
+        self,
+        server_round: int,
+        results: list[tuple[ClientProxy, FitRes]],
+        failures: list[tuple[ClientProxy, FitRes] | BaseException],
+    ) -> tuple[Parameters | None, dict[str, Scalar]]:
+        if not results:
+            return None, {}
+        if not self.accept_failures and failures:
+            return None, {}
+        dense_sums: dict[str, torch.Tensor] = {}
+        counts: dict[str, torch.Tensor] = {}
+        for _, res in results:
+            _, info = self.parameter_packer.unpack_parameters(res.parameters)
+            for name, vals, idx, shp in zip(info["names"], info["values"], info["indices"], info["shapes"]):
+                if name not in dense_sums:
+                    dense_sums[name] = torch.zeros(shp, dtype=torch.float32, device=vals.device)
+                    counts[name] = torch.zeros(shp, dtype=torch.float32, device=vals.device)
+                if idx.numel() == 0:
+                    continue
+                flat_idx = self._ravel(idx, shp)
+                dense_sums[name].reshape(-1).scatter_add_(0, flat_idx, vals.float())
+                counts[name].reshape(-1).scatter_add_(0, flat_idx, torch.ones_like(vals.float()))
+        names = list(dense_sums.keys())
+        values, indices, shapes = [], [], []
+        for name in names:
+            mask = counts[name] > 0
+            avg = torch.where(mask, dense_sums[name] / counts[name].clamp(min=1.0), torch.zeros_like(dense_sums[name]))
+            nz = mask.nonzero().t()
+            values.append(avg[mask])
+            indices.append(nz)
+            shapes.append(list(avg.shape))
+        packed = self.parameter_packer.pack_parameters(
+            Parameters([]), {"values": values, "indices": indices, "shapes": shapes, "names": names}
+        )
+        metrics = self.fit_metrics_aggregation_fn([(res.num_examples, res.metrics) for _, res in results])
+        return packed, metrics
+
+    @staticmethod
+    def _ravel(idx: torch.Tensor, shape: list[int]) -> torch.Tensor:
+        strides = torch.tensor(
+            [int(torch.tensor(shape[d + 1 :]).prod()) if d + 1 < len(shape) else 1 for d in range(len(shape))],
+            device=idx.device,
+        )
+        return (idx.long() * strides.unsqueeze(1)).sum(dim=0)

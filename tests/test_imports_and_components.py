@@ -1,0 +1,207 @@
+"""Every public subsystem imports and basic component behaviors hold."""
+import torch
+import torch.nn as nn
+
+
+def test_import_all_packages():
+    import fl4health_amd.clients as clients
+    import fl4health_amd.servers as servers
+    import fl4health_amd.strategies as strategies
+    import fl4health_amd.model_bases as model_bases
+    import fl4health_amd.model_bases.masked_layers as masked
+    import fl4health_amd.losses as losses
+    import fl4health_amd.metrics as metrics
+    import fl4health_amd.privacy as privacy
+    import fl4health_amd.checkpointing as ckpt
+    import fl4health_amd.reporting as reporting
+    import fl4health_amd.client_managers as cm
+    import fl4health_amd.parameter_exchange as pe
+
+    for mod in (clients, servers, strategies, model_bases, masked, losses, metrics, privacy, ckpt, reporting, cm, pe):
+        assert mod.__all__
+
+
+def test_masked_layers_convert_and_forward():
+    from fl4health_amd.model_bases.masked_layers import convert_to_masked_model, is_masked_module
+
+    model = nn.Sequential(nn.Conv2d(3, 4, 3, padding=1), nn.BatchNorm2d(4), nn.Flatten(), nn.Linear(4 * 8 * 8, 5))
+    masked = convert_to_masked_model(model)
+    assert is_masked_module(masked[0]) and is_masked_module(masked[1]) and is_masked_module(masked[3])
+    # weights frozen; scores trainable
+    assert not masked[0].weight.requires_grad
+    assert masked[0].weight_scores.requires_grad
+    x = torch.randn(2, 3, 8, 8)
+    out = masked(x)
+    assert out.shape == (2, 5)
+    out.sum().backward()
+    assert masked[0].weight_scores.grad is not None
+    assert masked[0].weight.grad is None
+
+
+def test_grad_sample_linear_matches_autograd():
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(6, 8), nn.ReLU(), nn.Linear(8, 3))
+    gsm = GradSampleModule(model)
+    x = torch.randn(4, 6)
+    y = torch.randint(0, 3, (4,))
+    loss = nn.functional.cross_entropy(gsm(x), y)
+    loss.backward()
+    # grad_sample[b] holds sample b's contribution to the total loss gradient,
+    # so the batch sum equals the autograd gradient exactly
+    for p in model.parameters():
+        assert hasattr(p, "grad_sample")
+        summed = p.grad_sample.sum(dim=0)
+        assert torch.allclose(summed, p.grad, atol=1e-5), f"max diff {(summed - p.grad).abs().max()}"
+
+
+def test_grad_sample_conv_matches_autograd():
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Conv2d(2, 3, 3, padding=1), nn.ReLU(), nn.Flatten(), nn.Linear(3 * 4 * 4, 2))
+    gsm = GradSampleModule(model)
+    x = torch.randn(5, 2, 4, 4)
+    y = torch.randint(0, 2, (5,))
+    loss = nn.functional.cross_entropy(gsm(x), y, reduction="sum")
+    loss.backward()
+    for name, p in model.named_parameters():
+        summed = p.grad_sample.sum(dim=0)
+        assert torch.allclose(summed, p.grad, atol=1e-4), f"{name}: {(summed - p.grad).abs().max()}"
+
+
+def test_dp_sgd_engine_runs():
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    model = nn.Linear(4, 2)
+    gsm = GradSampleModule(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    eng = DpSgdEngine(gsm, opt, noise_multiplier=0.5, clipping_bound=1.0, seed=3)
+    w0 = model.weight.detach().clone()
+    loss = gsm(torch.randn(8, 4)).sum()
+    eng.zero_grad()
+    loss.backward()
+    eng.step()
+    assert not torch.allclose(model.weight.detach(), w0)
+
+
+def test_moments_accountant_sane():
+    from fl4health_amd.privacy.moments_accountant import MomentsAccountant
+
+    acc = MomentsAccountant()
+    eps1 = acc.get_epsilon(0.01, 1.1, 1000, 1e-5)
+    eps2 = acc.get_epsilon(0.01, 1.1, 10000, 1e-5)
+    eps3 = acc.get_epsilon(0.01, 2.2, 1000, 1e-5)
+    assert 0 < eps1 < eps2  # more steps -> more privacy loss
+    assert eps3 < eps1  # more noise -> less privacy loss
+
+
+def test_mkmmd_loss_and_betas():
+    from fl4health_amd.losses.mkmmd_loss import MkMmdLoss
+
+    torch.manual_seed(0)
+    loss_fn = MkMmdLoss()
+    x = torch.randn(32, 8)
+    y_same = torch.randn(32, 8)
+    y_diff = torch.randn(32, 8) + 3.0
+    assert float(loss_fn(x, y_diff)) > float(loss_fn(x, y_same))
+    betas = loss_fn.optimize_betas(x, y_diff)
+    assert betas.shape == loss_fn.betas.shape
+    assert abs(float(betas.sum()) - 1.0) < 1e-4
+    assert (betas >= 0).all()
+
+
+def test_deep_mmd_loss_separates():
+    from fl4health_amd.losses.deep_mmd_loss import DeepMmdLoss
+
+    torch.manual_seed(0)
+    loss_fn = DeepMmdLoss("cpu", input_size=8, optimization_steps=2)
+    x = torch.randn(24, 8)
+    y_diff = torch.randn(24, 8) + 3.0
+    v_diff = float(loss_fn(x, y_diff))
+    loss_fn.training_loss = False
+    v_same = float(loss_fn(x, torch.randn(24, 8)))
+    assert v_diff > v_same
+
+
+def test_contrastive_losses():
+    from fl4health_amd.losses.contrastive_loss import MoonContrastiveLoss, NtXentLoss
+
+    torch.manual_seed(0)
+    f = torch.randn(8, 16)
+    pos = f + 0.01 * torch.randn(8, 16)
+    neg = -f.unsqueeze(0)
+    moon = MoonContrastiveLoss()
+    low = float(moon(f, pos.unsqueeze(0), neg))
+    high = float(moon(f, (-f).unsqueeze(0), (f + 0.01).unsqueeze(0)))
+    assert low < high
+    ntx = NtXentLoss()
+    assert float(ntx(f, pos)) < float(ntx(f, torch.randn(8, 16)))
+
+
+def test_fedpm_strategy_posterior():
+    from fl4health_amd.common import FitRes, Parameters
+    from fl4health_amd.strategies.fedpm import FedPm
+
+    class P:
+        def __init__(self, cid):
+            self.cid = cid
+
+    s = FedPm()
+    masks = [torch.tensor([1.0, 0.0, 1.0]), torch.tensor([1.0, 0.0, 0.0])]
+    results = [(P(str(i)), FitRes(Parameters([m]), 10, {})) for i, m in enumerate(masks)]
+    params, _ = s.aggregate_fit(1, results, [])
+    probs = params.tensors[0]
+    # alpha = 1 + [2,0,1]; lambda = 1 + [0,2,1]; mean = (alpha-1)/(alpha+lambda-2)
+    assert torch.allclose(probs, torch.tensor([1.0, 0.0, 0.5]))
+
+
+def test_sparse_coo_strategy_roundtrip():
+    from fl4health_amd.common import FitRes, Parameters
+    from fl4health_amd.parameter_exchange.packers import SparseCooParameterPacker
+    from fl4health_amd.strategies.fedavg_sparse_coo_tensor import FedAvgSparseCooTensor
+
+    class P:
+        def __init__(self, cid):
+            self.cid = cid
+
+    packer = SparseCooParameterPacker()
+    dense1 = torch.tensor([[1.0, 0.0], [0.0, 2.0]])
+    dense2 = torch.tensor([[3.0, 0.0], [0.0, 0.0]])
+    payloads = []
+    for d in (dense1, dense2):
+        idx = d.nonzero().t()
+        payloads.append(
+            packer.pack_parameters(
+                Parameters([]), {"values": [d[d != 0]], "indices": [idx], "shapes": [[2, 2]], "names": ["w"]}
+            )
+        )
+    s = FedAvgSparseCooTensor()
+    results = [(P(str(i)), FitRes(p, 5, {})) for i, p in enumerate(payloads)]
+    out, _ = s.aggregate_fit(1, results, [])
+    _, info = packer.unpack_parameters(out)
+    rec = torch.zeros(2, 2)
+    rec[tuple(info["indices"][0].long())] = info["values"][0]
+    # (1+3)/2 at [0,0]; 2/1 at [1,1]
+    assert torch.allclose(rec, torch.tensor([[2.0, 0.0], [0.0, 2.0]]))
+
+
+def test_feddg_ga_weight_update():
+    from fl4health_amd.strategies.feddg_ga import FairnessMetric, FairnessMetricType, FedDgGa
+
+    s = FedDgGa(fairness_metric=FairnessMetric(FairnessMetricType.LOSS))
+    s.num_rounds = 10
+    s.initial_adjustment_weight = 0.5
+    s.train_metrics = {
+        "0": {FairnessMetricType.LOSS.value: 1.0},
+        "1": {FairnessMetricType.LOSS.value: 1.0},
+    }
+    s.evaluation_metrics = {
+        "0": {FairnessMetricType.LOSS.value: 2.0},  # big gap
+        "1": {FairnessMetricType.LOSS.value: 1.0},  # no gap
+    }
+    s.update_weights_by_ga(1, ["0", "1"])
+    assert s.adjustment_weights["0"] > s.adjustment_weights["1"]
+    assert abs(sum(s.adjustment_weights.values()) - 1.0) < 1e-6
