@@ -69,3 +69,37 @@ def test_fedprox_simulation_on_gpu():
     assert all(torch.isfinite(torch.tensor(l)) for l in losses)
     # learning happened: loss moved
     assert losses[0] != losses[-1]
+
+
+@requires_gpu
+def test_dp_sgd_engine_on_gpu():
+    import torch.nn as nn
+
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(), nn.Flatten(), nn.Linear(8 * 8 * 8, 10)).cuda()
+    gsm = GradSampleModule(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    eng = DpSgdEngine(gsm, opt, noise_multiplier=0.0, clipping_bound=1e9, seed=1)
+    x = torch.randn(16, 3, 8, 8, device="cuda")
+    y = torch.randint(0, 10, (16,), device="cuda")
+
+    # with no clipping and no noise, DP-SGD step == plain SGD step
+    import copy
+
+    ref_model = copy.deepcopy(model)
+    ref_opt = torch.optim.SGD(ref_model.parameters(), lr=0.05)
+    loss_ref = nn.functional.cross_entropy(ref_model(x), y)
+    ref_opt.zero_grad()
+    loss_ref.backward()
+    ref_opt.step()
+
+    loss = nn.functional.cross_entropy(gsm(x), y)
+    eng.zero_grad()
+    loss.backward()
+    eng.step()
+    torch.cuda.synchronize()
+    for p, pr in zip(model.parameters(), ref_model.parameters()):
+        assert torch.allclose(p, pr, atol=1e-4), f"max diff {(p - pr).abs().max()}"
