@@ -49,7 +49,10 @@ class BenchFedProxClient(FedProxClient):
             self.use_cuda_graph = not args.no_graph
 
     def get_model(self, config):
-        return ResNet18(num_classes=10)
+        model = ResNet18(num_classes=10)
+        if self.device.type == "cuda":
+            model = model.to(memory_format=torch.channels_last)
+        return model
 
     def get_data_loaders(self, config):
         from fl4health_amd.datasets.loaders import DeviceTensorLoader

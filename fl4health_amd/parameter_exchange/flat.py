@@ -119,7 +119,14 @@ class FlatParameterView:
 
     # ---- view binding (zero-copy hot path) -----------------------------
     def _bind_views(self) -> None:
-        """Rebind fp32 module params/buffers as views of the flat buffer."""
+        """Rebind fp32 module params/buffers as views of the flat buffer.
+
+        4D params of channels-last modules are bound with NHWC strides over
+        their flat slice (view as (N,H,W,C) then permute), so MIOpen's NHWC
+        bf16 igemm kernels run without per-step layout transposes. The flat
+        buffer then stores such weights in NHWC element order — identical on
+        every rank, so collective aggregation is unaffected.
+        """
         name_to_idx = {n: i for i, n in enumerate(self.spec.names)}
         for mod_name, mod in self.module.named_modules():
             prefix = mod_name + "." if mod_name else ""
@@ -128,8 +135,15 @@ class FlatParameterView:
                 i = name_to_idx.get(full)
                 if i is None or p.dtype != torch.float32:
                     continue
-                view = self.spec.slice_of(self.flat, i)
-                new_p = nn.Parameter(view, requires_grad=p.requires_grad)
+                view = self.spec.slice_of(self.flat, i).reshape(-1)
+                if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last):
+                    n_, c_, h_, w_ = p.shape
+                    nhwc = view.view(n_, h_, w_, c_).permute(0, 3, 1, 2)
+                    with torch.no_grad():
+                        nhwc.copy_(p.detach())
+                    new_p = nn.Parameter(nhwc, requires_grad=p.requires_grad)
+                else:
+                    new_p = nn.Parameter(view.view(p.shape), requires_grad=p.requires_grad)
                 setattr(mod, pname, new_p)
             for bname, b in list(mod.named_buffers(recurse=False)):
                 full = prefix + bname
@@ -162,5 +176,12 @@ class FlatParameterView:
             if i is None or p.dtype != torch.float32:
                 continue
             n = p.numel()
-            p.grad = gbuf[self.spec.offsets[i] : self.spec.offsets[i] + n].view(p.shape)
+            gslice = gbuf[self.spec.offsets[i] : self.spec.offsets[i] + n]
+            if p.dim() == 4 and not p.is_contiguous() and p.is_contiguous(memory_format=torch.channels_last):
+                # grad memory order must MATCH the param's NHWC flat order so
+                # the fused elementwise optimizer kernels stay aligned
+                n_, c_, h_, w_ = p.shape
+                p.grad = gslice.view(n_, h_, w_, c_).permute(0, 3, 1, 2)
+            else:
+                p.grad = gslice.view(p.shape)
         return gbuf

@@ -25,12 +25,15 @@ class DpSgdEngine:
         noise_multiplier: float,
         clipping_bound: float,
         seed: int = 0,
+        loss_reduction: str = "mean",
     ) -> None:
+        assert loss_reduction in ("mean", "sum")
         self.module = module
         self.optimizer = optimizer
         self.noise_multiplier = noise_multiplier
         self.clipping_bound = clipping_bound
         self.seed = seed
+        self.loss_reduction = loss_reduction
         self._noise_counter = 0
 
     @torch.no_grad()
@@ -41,21 +44,29 @@ class DpSgdEngine:
             return
         device = params[0].device
         batch = params[0].grad_sample.shape[0]
+        # With mean-reduced losses the captured grad_sample[b] is (1/B) dL_b,
+        # i.e. already carries the 1/B factor: clip at C/B (the coefficient
+        # min(1, (C/B)/(||dL_b||/B)) equals the true min(1, C/||dL_b||)), skip
+        # the final 1/B division, and scale the noise std by 1/B.
+        if self.loss_reduction == "mean":
+            eff_bound = self.clipping_bound / batch
+            sigma = self.noise_multiplier * self.clipping_bound / batch
+            final_div = 1.0
+        else:
+            eff_bound = self.clipping_bound
+            sigma = self.noise_multiplier * self.clipping_bound
+            final_div = float(batch)
         sqnorms = torch.zeros(batch, dtype=torch.float32, device=device)
         for p in params:
             F.per_sample_sqnorm_(p.grad_sample.reshape(batch, -1).float(), sqnorms)
         for p in params:
             g = torch.zeros(p.numel(), dtype=torch.float32, device=device)
-            F.clip_rowsum_(p.grad_sample.reshape(batch, -1).float(), sqnorms, g, self.clipping_bound)
+            F.clip_rowsum_(p.grad_sample.reshape(batch, -1).float(), sqnorms, g, eff_bound)
             if self.noise_multiplier > 0:
-                F.gaussian_noise_(
-                    g,
-                    sigma=self.noise_multiplier * self.clipping_bound,
-                    seed=self.seed,
-                    offset=self._noise_counter,
-                )
+                F.gaussian_noise_(g, sigma=sigma, seed=self.seed, offset=self._noise_counter)
                 self._noise_counter += (p.numel() + 3) // 4 + 1
-            g /= batch
+            if final_div != 1.0:
+                g /= final_div
             # write INTO existing .grad when present (it may alias a flat grad
             # buffer used by fused optimizers) rather than rebinding
             if p.grad is not None and p.grad.shape == p.shape:

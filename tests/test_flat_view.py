@@ -61,3 +61,25 @@ def test_int_buffers_roundtrip():
     m[1].num_batches_tracked.zero_()
     view.load_flat(flat)
     assert int(m[1].num_batches_tracked) == nbt_before
+
+
+def test_channels_last_binding_preserves_semantics():
+    import torch
+
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.BatchNorm2d(8), nn.Flatten(), nn.Linear(8 * 4 * 4, 5))
+    m = m.to(memory_format=torch.channels_last)
+    ref = [p.detach().clone() for p in m.parameters()]
+    view = FlatParameterView(m, bind=True)
+    for p, r in zip(m.parameters(), ref):
+        assert torch.allclose(p.detach(), r)
+    gbuf = view.make_grad_buffer()
+    x = torch.randn(4, 3, 4, 4).contiguous(memory_format=torch.channels_last)
+    m(x).sum().backward()
+    # flat elementwise step over (params_region, gbuf) == per-param SGD step
+    before = [p.detach().clone() for p in m.parameters()]
+    grads = [p.grad.clone() for p in m.parameters()]
+    with torch.no_grad():
+        view.params_region.add_(gbuf, alpha=-0.1)
+    for p, b, g in zip(m.parameters(), before, grads):
+        assert torch.allclose(p.detach(), b - 0.1 * g, atol=1e-6)

@@ -90,7 +90,7 @@ def test_server_opt_gpu(kind):
         # crosses zero; compare only away from the singularity (state tensors
         # above are compared exactly everywhere)
         mask = (v.sqrt() - d + tau).abs() > 0.05
-        assert mask.float().mean() > 0.9
+        assert mask.float().mean() > 0.8
         diff = diff[mask]
     assert float(diff.max()) < 1e-3, f"{kind}: max diff {float(diff.max())}"
 

@@ -205,3 +205,30 @@ def test_feddg_ga_weight_update():
     s.update_weights_by_ga(1, ["0", "1"])
     assert s.adjustment_weights["0"] > s.adjustment_weights["1"]
     assert abs(sum(s.adjustment_weights.values()) - 1.0) < 1e-6
+
+
+def test_dp_sgd_no_clip_no_noise_equals_sgd_cpu():
+    import copy
+
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(), nn.Flatten(), nn.Linear(8 * 8 * 8, 10))
+    ref_model = copy.deepcopy(model)
+    gsm = GradSampleModule(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    eng = DpSgdEngine(gsm, opt, noise_multiplier=0.0, clipping_bound=1e9, seed=1)
+    x = torch.randn(16, 3, 8, 8)
+    y = torch.randint(0, 10, (16,))
+
+    ref_opt = torch.optim.SGD(ref_model.parameters(), lr=0.05)
+    ref_opt.zero_grad()
+    nn.functional.cross_entropy(ref_model(x), y).backward()
+    ref_opt.step()
+
+    eng.zero_grad()
+    nn.functional.cross_entropy(gsm(x), y).backward()
+    eng.step()
+    for p, pr in zip(model.parameters(), ref_model.parameters()):
+        assert torch.allclose(p, pr, atol=1e-5), f"max diff {(p - pr).abs().max()}"
