@@ -205,6 +205,17 @@ class BasicClient:
 
     def set_parameters(self, parameters: Parameters, config: Config, fitting_round: bool) -> None:
         assert self.initialized
+        # round-0 initialization handshake ships FULL weights regardless of the
+        # configured (possibly partial) exchanger (reference basic_client.py:179):
+        # a payload without layer metadata whose size matches the full flat
+        # buffer is loaded wholesale
+        if (
+            "layer_names" not in parameters.meta
+            and len(parameters.tensors) >= 1
+            and parameters.tensors[0].numel() == self.flat_view.spec.total
+        ):
+            self.flat_view.load_flat(parameters.tensors[0])
+            return
         self.parameter_exchanger.pull_parameters(parameters, self.model, config)
 
     # ------------------------------------------------------------------
@@ -409,6 +420,11 @@ class BasicClient:
                 self.reports_manager.report(report_data, current_round, local_epoch, self.total_steps)
                 self.total_steps += 1
                 steps_this_round += 1
+                if self.early_stopper is not None and self.early_stopper.should_stop(self.total_steps):
+                    self.early_stopper.load_snapshot()
+                    metrics = self.train_metric_manager.compute()
+                    loss_dict = self.train_loss_meter.compute()
+                    return loss_dict, metrics
             metrics = self.train_metric_manager.compute()
             loss_dict = self.train_loss_meter.compute()
             self._log_results(loss_dict, metrics, current_round, local_epoch)
@@ -438,6 +454,9 @@ class BasicClient:
             report_data.update({"fit_step": self.total_steps})
             self.reports_manager.report(report_data, current_round, None, self.total_steps)
             self.total_steps += 1
+            if self.early_stopper is not None and self.early_stopper.should_stop(self.total_steps):
+                self.early_stopper.load_snapshot()
+                break
         loss_dict = self.train_loss_meter.compute()
         metrics = self.train_metric_manager.compute()
         self._log_results(loss_dict, metrics, current_round)

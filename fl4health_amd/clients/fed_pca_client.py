@@ -56,7 +56,11 @@ class FedPCAClient(BasicClient):
         return Parameters([components, singular_values]), self.num_train_samples, {}
 
     def get_parameters(self, config: Config) -> Parameters:
-        assert self.model.principal_components is not None
+        self.maybe_setup_client(config)
+        if self.model.principal_components is None:
+            # initialization poll before any local SVD: run it now
+            x = self._data_matrix(self.train_loader)
+            self.model(x, center_data=bool(config.get("center_data", True)))
         return Parameters([self.model.principal_components, self.model.singular_values])
 
     def set_parameters(self, parameters: Parameters, config: Config, fitting_round: bool) -> None:

@@ -1,0 +1,81 @@
+"""FENDA + Ditto client (reference fl4health/clients/fenda_ditto_client.py:21-345):
+a FENDA personal model trained alongside a Ditto global model, with the
+drift penalty tying the FENDA GLOBAL feature extractor to the aggregated
+global model's feature extractor."""
+from __future__ import annotations
+
+import copy
+
+import torch
+
+from fl4health_amd.clients.basic_client import BasicClient, TorchPredType
+from fl4health_amd.common import Config, Parameters
+from fl4health_amd.losses.weight_drift_loss import WeightDriftLoss
+from fl4health_amd.model_bases.fenda_base import FendaModel
+from fl4health_amd.model_bases.sequential_split_models import SequentiallySplitExchangeBaseModel
+from fl4health_amd.parameter_exchange.exchangers import FullParameterExchangerWithPacking
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.parameter_exchange.packers import ParameterPackerAdaptiveConstraint
+from fl4health_amd.utils.losses import TrainingLosses
+
+
+class FendaDittoClient(BasicClient):
+    def __init__(self, *args, lam: float = 1.0, freeze_global_feature_extractor: bool = False, **kwargs) -> None:
+        super().__init__(*args, **kwargs)
+        self.lam = lam
+        self.freeze_global_feature_extractor = freeze_global_feature_extractor
+        self.model: FendaModel
+        self.global_model: SequentiallySplitExchangeBaseModel
+        self.drift_loss = WeightDriftLoss()
+        self._anchor_weights: list[torch.Tensor] | None = None
+        self._vanilla_loss = 0.0
+
+    def get_global_model(self, config: Config) -> SequentiallySplitExchangeBaseModel:
+        raise NotImplementedError("user must supply the Ditto global model architecture")
+
+    def get_parameter_exchanger(self, config: Config) -> FullParameterExchangerWithPacking:
+        return FullParameterExchangerWithPacking(ParameterPackerAdaptiveConstraint())
+
+    def setup_client(self, config: Config) -> None:
+        super().setup_client(config)
+        self.global_model = self.get_global_model(config).to(self.device)
+        self.global_flat_view = FlatParameterView(self.global_model, bind=True)
+        assert "global" in self.optimizers and "local" in self.optimizers
+
+    def set_parameters(self, parameters: Parameters, config: Config, fitting_round: bool) -> None:
+        assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
+        model_params, self.lam = self.parameter_exchanger.unpack_parameters(parameters)
+        self.global_flat_view.load_flat(model_params.tensors[0])
+        if fitting_round:
+            # FENDA global extractor initialized from / anchored to the Ditto
+            # global model's feature extractor
+            src = dict(self.global_model.base_module.state_dict())
+            if self.current_server_round <= 1 or self.freeze_global_feature_extractor:
+                self.model.second_feature_extractor.load_state_dict(src)
+            self._anchor_weights = [p.detach().clone() for p in self.global_model.base_module.parameters()]
+
+    def get_parameters(self, config: Config) -> Parameters:
+        assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
+        model_params = self.parameter_exchanger.push_parameters(self.global_model, config=config)
+        return self.parameter_exchanger.pack_parameters(model_params, self._vanilla_loss)
+
+    def train_step(self, input, target):
+        self.set_optimizer_zero_grad()
+        global_out = self.global_model(input)
+        global_pred = global_out[0]["prediction"] if isinstance(global_out, tuple) else global_out
+        global_loss = self.criterion(global_pred, target)
+        preds, features = self.model(input)
+        local_loss = self.criterion(preds["prediction"], target)
+        penalty = self.drift_loss(self.model.second_feature_extractor, self._anchor_weights, self.lam)
+        total = local_loss + penalty
+        global_loss.backward()
+        total.backward()
+        self.optimizers["global"].step()
+        self.optimizers["local"].step()
+        losses = TrainingLosses(
+            backward={"backward": total.detach()},
+            additional_losses={"global_loss": global_loss.detach(), "local_loss": local_loss.detach(), "penalty_loss": penalty.detach()},
+        )
+        self._vanilla_loss = float(global_loss.detach())
+        all_preds: TorchPredType = {**preds, "global": global_pred}
+        return losses, all_preds

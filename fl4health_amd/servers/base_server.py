@@ -265,7 +265,20 @@ class FlServer:
 
     def _load_server_state(self) -> bool:
         loaded = self.checkpoint_and_state_module.maybe_load_state(self, f"server_{self.server_name}_state.pt")
-        return loaded is not None
+        if loaded is None:
+            return False
+        # rebuild server parameters from the saved (hydrated) model weights
+        model = self.checkpoint_and_state_module.model
+        if model is not None and "model" in loaded:
+            state = loaded["model"]
+            if isinstance(state, dict):
+                model.load_state_dict(state)
+            restored = self.checkpoint_and_state_module.parameter_exchanger.push_parameters(model)
+            if self.parameters is not None and len(self.parameters.tensors) > len(restored.tensors):
+                # re-attach strategy aux payloads (mu, variates...) to the restored weights
+                restored.tensors = restored.tensors + self.parameters.tensors[len(restored.tensors):]
+            self.parameters = restored
+        return True
 
     def shutdown(self) -> None:
         self.reports_manager.report({"shutdown": str(datetime.datetime.now())})

@@ -77,21 +77,23 @@ def test_server_opt_gpu(kind):
     xg = x.cuda()
     m, v, d = torch.zeros(n), torch.zeros(n), torch.zeros(n)
     mg, vg, dg = (t.cuda() for t in (m.clone(), v.clone(), d.clone()))
+    # flash's update m/(sqrt(v)-d+tau) is discontinuous where the denominator
+    # crosses zero: track, per iteration, which elements stayed safely away
+    # from the singularity and compare x only there (the moment state tensors
+    # are compared tightly everywhere).
+    safe = torch.ones(n, dtype=torch.bool)
     for _ in range(3):
         delta = torch.randn(n)
         F.server_opt_step_(x, delta, m, v, d, kind=kind, lr=0.1, tau=tau)
         F.server_opt_step_(xg, delta.cuda(), mg, vg, dg, kind=kind, lr=0.1, tau=tau)
+        safe &= (v.sqrt() - d + tau).abs() > 0.05
     assert torch.allclose(mg.cpu(), m, atol=1e-5)
     assert torch.allclose(vg.cpu(), v, atol=1e-5)
     assert torch.allclose(dg.cpu(), d, atol=1e-5)
     diff = (xg.cpu() - x).abs()
     if kind == "flash":
-        # the update m/(sqrt(v)-d+tau) is discontinuous where the denominator
-        # crosses zero; compare only away from the singularity (state tensors
-        # above are compared exactly everywhere)
-        mask = (v.sqrt() - d + tau).abs() > 0.05
-        assert mask.float().mean() > 0.8
-        diff = diff[mask]
+        assert safe.float().mean() > 0.5
+        diff = diff[safe]
     assert float(diff.max()) < 1e-3, f"{kind}: max diff {float(diff.max())}"
 
 
