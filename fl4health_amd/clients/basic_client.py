@@ -39,6 +39,7 @@ from fl4health_amd.reporting.reports_manager import ReportsManager
 from fl4health_amd.utils.config import narrow_dict_type
 from fl4health_amd.utils.losses import EvaluationLosses, LossMeter, LossMeterType, TrainingLosses
 from fl4health_amd.utils.random import generate_hash
+from fl4health_amd.utils.tracing import trace_range
 
 log = logging.getLogger(__name__)
 
@@ -411,7 +412,8 @@ class BasicClient:
             for input, target in self.train_loader:
                 self.update_before_step(steps_this_round, current_round)
                 input, target = self._move_to_device(input), self._move_to_device(target)
-                losses, preds = self._dispatch_train_step(input, target)
+                with trace_range("train_step"):
+                    losses, preds = self._dispatch_train_step(input, target)
                 self.train_loss_meter.update(losses)
                 self.train_metric_manager.update(preds, target)
                 self.update_after_step(steps_this_round, current_round)

@@ -38,6 +38,7 @@ from fl4health_amd.common import (
     Parameters,
 )
 from fl4health_amd.metrics.metric_aggregation import metric_aggregation
+from fl4health_amd.utils.tracing import trace_range
 
 log = logging.getLogger(__name__)
 
@@ -236,6 +237,7 @@ class DistributedRuntime:
                 scales = strategy.collective_scales(
                     fit_res.num_examples, total_examples, cohort_size, len(fit_res.parameters.tensors)
                 )
+                # pre-scale + concat: one comm buffer per round (K1 pre-scaling)
                 buf = torch.cat(
                     [
                         (t.reshape(-1).to(self.comm_device, torch.float32) * s)
@@ -248,7 +250,8 @@ class DistributedRuntime:
                 if self.rank in ok and strategy is None:
                     raise RuntimeError("collective aggregation requires a replicated strategy object on every rank")
                 buf = torch.zeros(total_numel, dtype=torch.float32, device=self.comm_device)
-            dist.all_reduce(buf, op=dist.ReduceOp.SUM)
+            with trace_range("fl_allreduce_aggregate"):
+                dist.all_reduce(buf, op=dist.ReduceOp.SUM)
             if self.rank == 0:
                 tensors = []
                 off = 0
