@@ -399,3 +399,82 @@ def test_smoke_tabular_feature_alignment():
     )
     hist = run_simulation(server, clients, num_rounds=2)
     assert len(hist.losses_distributed) == 2
+
+
+def test_smoke_nnunet_segmentation():
+    from fl4health_amd.clients.nnunet_client import NnunetClient
+    from fl4health_amd.servers.nnunet_server import NnunetServer
+    from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+
+    set_all_random_seeds(42)
+    cfg = {
+        "n_server_rounds": 1, "batch_size": 1, "patch_size": [16, 16, 16],
+        "num_classes": 2, "base_channels": 4, "num_levels": 2,
+        "n_train_volumes": 2, "n_val_volumes": 1,
+    }
+    clients = [NnunetClient(device="cpu", client_name=f"seg{i}") for i in range(2)]
+    strategy = BasicFedAvg(on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 1, **cfg})
+    server = NnunetServer(SimpleClientManager(), cfg, strategy)
+    hist = run_simulation(server, clients, num_rounds=1)
+    assert len(hist.losses_distributed) == 1
+    assert server.nnunet_plans is not None
+
+
+def test_smoke_bert_moon_lora():
+    from torch.utils.data import DataLoader, TensorDataset
+
+    from fl4health_amd.clients.moon_client import MoonClient
+    from fl4health_amd.models.bert import BertMoonModel, synthetic_agnews_batch
+    from fl4health_amd.models.lora import apply_lora, get_lora_parameter_names
+    from fl4health_amd.parameter_exchange.exchangers import FixedLayerExchanger
+    from fl4health_amd.strategies.fedavg_dynamic_layer import FedAvgDynamicLayer
+
+    set_all_random_seeds(42)
+
+    class Client(MoonClient):
+        def __init__(self, seed, **kw):
+            super().__init__(**kw)
+            self.seed = seed
+
+        def get_model(self, config):
+            model = BertMoonModel(num_classes=4, small=True)
+            return apply_lora(model, ("query", "value"), r=4)
+
+        def get_parameter_exchanger(self, config):
+            # LoRA/PEFT subset exchange (+ classification head)
+            names = get_lora_parameter_names(self.model) + [
+                n for n in self.model.state_dict() if n.startswith("head.")
+            ]
+            return FixedLayerExchanger(names)
+
+        def get_data_loaders(self, config):
+            ids, mask, y = synthetic_agnews_batch(24, seq_len=32, vocab=4096, seed=self.seed)
+            train = TensorDataset(ids, mask, y)
+
+            def collate(batch):
+                i, m, t = zip(*batch)
+                return {"input_ids": torch.stack(i), "attention_mask": torch.stack(m)}, torch.stack(t)
+
+            return (
+                DataLoader(train, batch_size=8, collate_fn=collate),
+                DataLoader(train, batch_size=8, collate_fn=collate),
+            )
+
+        def get_optimizer(self, config):
+            return torch.optim.AdamW([p for p in self.model.parameters() if p.requires_grad], lr=1e-4)
+
+        def get_criterion(self, config):
+            return nn.CrossEntropyLoss()
+
+        def predict(self, input):
+            preds, features = MoonClient.predict(self, input)
+            return preds, features
+
+    clients = [Client(seed=i, metrics=[Accuracy()], device="cpu") for i in range(2)]
+    strategy = FedAvgDynamicLayer(on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 1})
+    server = FlServer(SimpleClientManager(), {"n_server_rounds": 1, "batch_size": 8}, strategy)
+    hist = run_simulation(server, clients, num_rounds=1)
+    assert len(hist.losses_distributed) == 1
+    # only adapter + head weights crossed the boundary
+    p = clients[0].get_parameters({"current_server_round": 1})
+    assert all(("lora" in n) or n.startswith("head.") for n in p.meta["layer_names"])
