@@ -1,0 +1,66 @@
+"""Shared helpers for the example entry points.
+
+Each example runs either:
+- in-process simulation (default; deterministic, CPU-friendly):
+    python -m examples.<name>.run --n_clients 2 --rounds 3
+- one-rank-per-GPU distributed over RCCL/xGMI:
+    torchrun --nproc-per-node N -m examples.<name>.run --distributed
+"""
+from __future__ import annotations
+
+import argparse
+
+import torch
+
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.common import Parameters
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed, run_simulation
+from fl4health_amd.utils.random import set_all_random_seeds
+
+
+def example_argparser(description: str) -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(description=description)
+    p.add_argument("--n_clients", type=int, default=2)
+    p.add_argument("--rounds", type=int, default=3)
+    p.add_argument("--local_steps", type=int, default=5)
+    p.add_argument("--batch_size", type=int, default=32)
+    p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--distributed", action="store_true", help="run one client per rank (torchrun)")
+    p.add_argument("--config_path", type=str, default=None, help="optional YAML overriding defaults")
+    return p
+
+
+def load_overrides(args) -> dict:
+    if args.config_path:
+        from fl4health_amd.utils.config import load_config
+
+        return load_config(args.config_path)
+    return {}
+
+
+def initial_parameters(model_fn) -> Parameters:
+    return Parameters([FlatParameterView(model_fn()).flat.clone()])
+
+
+def launch(args, server_factory, client_factory, strategy_factory=None):
+    set_all_random_seeds(args.seed)
+    if args.distributed:
+        hist = run_distributed(server_factory, lambda rank, world: client_factory(rank), args.rounds, strategy_factory)
+        if hist is not None:
+            _report(hist)
+        return hist
+    clients = [client_factory(i) for i in range(args.n_clients)]
+    server = server_factory()
+    hist = run_simulation(server, clients, num_rounds=args.rounds)
+    _report(hist)
+    return hist
+
+
+def _report(hist) -> None:
+    print("[SUMMARY] aggregated val losses by round:")
+    for r, loss in hist.losses_distributed:
+        print(f"  round {r}: {loss:.4f}")
+    for key, vals in hist.metrics_distributed.items():
+        print(f"[SUMMARY] {key}: {[(r, round(float(v), 4)) for r, v in vals]}")

@@ -1,0 +1,59 @@
+"""SCAFFOLD example (capability of reference examples/scaffold_example):
+control variates over RCCL, fused variate-corrected SGD kernel."""
+from __future__ import annotations
+
+import torch
+
+from examples.common import example_argparser, initial_parameters, launch
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.clients.scaffold_client import ScaffoldClient
+from fl4health_amd.datasets.synthetic import synthetic_cifar_loaders
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.models.cnn import SmallCnn
+from fl4health_amd.optimizers import FlatScaffoldSGD
+from fl4health_amd.servers.scaffold_server import ScaffoldServer
+from fl4health_amd.strategies.scaffold import Scaffold
+
+
+class Client(ScaffoldClient):
+    def __init__(self, seed: int, args, **kw) -> None:
+        super().__init__(**kw)
+        self.seed = seed
+        self.args = args
+
+    def get_model(self, config):
+        return SmallCnn()
+
+    def get_data_loaders(self, config):
+        return synthetic_cifar_loaders(n_train=1024, n_val=256, batch_size=self.args.batch_size, seed=self.seed)
+
+    def get_optimizer(self, config):
+        return FlatScaffoldSGD(self.flat_view, lr=0.05)
+
+    def get_criterion(self, config):
+        return torch.nn.CrossEntropyLoss()
+
+
+def main() -> None:
+    args = example_argparser("SCAFFOLD example").parse_args()
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    def strategy_factory():
+        return Scaffold(
+            initial_parameters=initial_parameters(SmallCnn),
+            on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": args.local_steps},
+        )
+
+    def server_factory():
+        return ScaffoldServer(
+            SimpleClientManager(), {"n_server_rounds": args.rounds, "batch_size": args.batch_size}, strategy_factory()
+        )
+
+    def client_factory(cid: int):
+        return Client(cid, args, metrics=[Accuracy()], device=device)
+
+    launch(args, server_factory, client_factory, strategy_factory)
+
+
+if __name__ == "__main__":
+    main()
