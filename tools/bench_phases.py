@@ -1,9 +1,12 @@
 """Phase-level timing of one FL round (world=1) to locate round overhead."""
+import os
+import sys
 import time
 
 import torch
 
-import bench as B
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import bench as B  # noqa: E402
 
 
 def main() -> None:
