@@ -49,7 +49,9 @@ class BenchFedProxClient(FedProxClient):
     def get_model(self, config):
         model = ResNet18(num_classes=10)
         if self.device.type == "cuda":
-            model = model.to(memory_format=torch.channels_last)
+            from fl4health_amd.ops.batchnorm import convert_batchnorm_to_cdna
+
+            model = convert_batchnorm_to_cdna(model.to(memory_format=torch.channels_last))
         return model
 
     def get_data_loaders(self, config):

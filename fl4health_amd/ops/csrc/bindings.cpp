@@ -24,6 +24,11 @@ void launch_clip_rowsum(const float*, const float*, float*, float, int64_t, int6
 void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, int64_t,
                       hipStream_t);
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
+void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
+                   float*, float, float, int64_t, int, int, int, hipStream_t);
+void launch_bn_bwd(const void*, const void*, void*, float*, const float*, const float*,
+                   const float*, float*, float*, float*, float*, int64_t, int, int, int,
+                   hipStream_t);
 }
 
 namespace {
@@ -180,9 +185,66 @@ torch::Tensor weighted_sum_rows(torch::Tensor stack, torch::Tensor w) {
   return out;
 }
 
+constexpr int BN_GROUPS = 64;
+
+int bn_dtype_of(const torch::Tensor& t) {
+  if (t.scalar_type() == torch::kBFloat16) return 1;
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, "bn ops support fp32/bf16");
+  return 0;
+}
+
+// x: NHWC-flattened [R, C] contiguous. Returns (y, save_mean, save_invstd).
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Tensor> gamma,
+                                        c10::optional<torch::Tensor> beta,
+                                        c10::optional<torch::Tensor> running_mean,
+                                        c10::optional<torch::Tensor> running_var, double momentum,
+                                        double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2, "x must be [R, C] contiguous");
+  int64_t R = x.size(0);
+  int C = (int)x.size(1);
+  int dtype = bn_dtype_of(x);
+  auto fopts = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
+  auto y = torch::empty_like(x);
+  auto partial = torch::empty({2LL * BN_GROUPS * C}, fopts);
+  auto mean = torch::empty({C}, fopts);
+  auto invstd = torch::empty({C}, fopts);
+  launch_bn_fwd(
+      x.data_ptr(), y.data_ptr(), partial.data_ptr<float>(), mean.data_ptr<float>(),
+      invstd.data_ptr<float>(),
+      gamma.has_value() ? gamma->data_ptr<float>() : nullptr,
+      beta.has_value() ? beta->data_ptr<float>() : nullptr,
+      running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
+      running_var.has_value() ? running_var->data_ptr<float>() : nullptr, (float)momentum,
+      (float)eps, R, C, BN_GROUPS, dtype, stream());
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
+                                  torch::Tensor invstd, c10::optional<torch::Tensor> gamma) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous() && x.sizes() == dy.sizes());
+  int64_t R = x.size(0);
+  int C = (int)x.size(1);
+  int dtype = bn_dtype_of(x);
+  auto fopts = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
+  auto dx = torch::empty_like(x);
+  auto partial = torch::empty({2LL * BN_GROUPS * C}, fopts);
+  auto sum_dy = torch::empty({C}, fopts);
+  auto sum_dy_xhat = torch::empty({C}, fopts);
+  auto dgamma = torch::empty({C}, fopts);
+  auto dbeta = torch::empty({C}, fopts);
+  launch_bn_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), partial.data_ptr<float>(),
+                mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                gamma.has_value() ? gamma->data_ptr<float>() : nullptr, sum_dy.data_ptr<float>(),
+                sum_dy_xhat.data_ptr<float>(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                R, C, BN_GROUPS, dtype, stream());
+  return {dx, dgamma, dbeta};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_fwd_train", &bn_fwd_train, "NHWC batchnorm training forward");
+  m.def("bn_bwd", &bn_bwd, "NHWC batchnorm backward");
   m.def("axpby_", &axpby_, "y = a*x + b*y (in-place)");
   m.def("prox_sgd_step_", &prox_sgd_step_, "fused proximal SGD step");
   m.def("scaffold_sgd_step_", &scaffold_sgd_step_, "fused SCAFFOLD-corrected SGD step");

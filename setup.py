@@ -22,6 +22,7 @@ setup(
             sources=[
                 os.path.join(CSRC, "bindings.cpp"),
                 os.path.join(CSRC, "flat_ops.hip"),
+                os.path.join(CSRC, "bn_ops.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3"],

@@ -183,3 +183,46 @@ def test_weighted_sum_rows_gpu():
     out_c = F.weighted_sum_rows(stack, w)
     out_g = F.weighted_sum_rows(stack.cuda(), w.cuda())
     assert torch.allclose(out_g.cpu(), out_c, atol=1e-4)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_cdna_batchnorm_matches_torch(dtype):
+    from fl4health_amd.ops.batchnorm import CdnaBatchNorm2d
+
+    torch.manual_seed(0)
+    n, c, h, w = 16, 64, 16, 16
+    x = torch.randn(n, c, h, w, device="cuda").to(dtype).contiguous(memory_format=torch.channels_last)
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+
+    ref = torch.nn.BatchNorm2d(c).cuda()
+    ours = CdnaBatchNorm2d(c).cuda()
+    ours.load_state_dict(ref.state_dict())
+    ref.train()
+    ours.train()
+
+    y_ref = ref(x1.float())
+    y_ours = ours(x2)
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(y_ours.float(), y_ref, atol=tol), f"fwd max diff {(y_ours.float()-y_ref).abs().max()}"
+    assert torch.allclose(ours.running_mean, ref.running_mean, atol=tol)
+    assert torch.allclose(ours.running_var, ref.running_var, atol=tol)
+    assert int(ours.num_batches_tracked) == int(ref.num_batches_tracked)
+
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y_ours.backward(g.to(dtype))
+    assert torch.allclose(x2.grad.float(), x1.grad, atol=tol * 4), f"dx max diff {(x2.grad.float()-x1.grad).abs().max()}"
+    assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=tol * 10, rtol=1e-2)
+    assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=tol * 10, rtol=1e-2)
+
+
+@requires_gpu
+def test_cdna_batchnorm_deterministic():
+    from fl4health_amd.ops import functional as F
+
+    x = torch.randn(100000, 64, device="cuda")
+    y1, m1, v1 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5)
+    y2, m2, v2 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5)
+    assert torch.equal(m1, m2) and torch.equal(v1, v2) and torch.equal(y1, y2)
