@@ -185,7 +185,16 @@ torch::Tensor weighted_sum_rows(torch::Tensor stack, torch::Tensor w) {
   return out;
 }
 
-constexpr int BN_GROUPS = 64;
+// Row-group count: a pure function of the shape (deterministic), sized so the
+// grid covers all 256 CUs many times over (~4096 blocks regardless of C).
+int bn_groups(int64_t R, int C) {
+  int cblocks = (C + 127) / 128;
+  int64_t g = 4096 / cblocks;
+  if (g > (R + 3) / 4) g = (R + 3) / 4;  // at least ~4 rows per group
+  if (g < 1) g = 1;
+  if (g > 8192) g = 8192;
+  return (int)g;
+}
 
 int bn_dtype_of(const torch::Tensor& t) {
   if (t.scalar_type() == torch::kBFloat16) return 1;
@@ -203,9 +212,10 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
   int64_t R = x.size(0);
   int C = (int)x.size(1);
   int dtype = bn_dtype_of(x);
+  int G = bn_groups(R, C);
   auto fopts = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
   auto y = torch::empty_like(x);
-  auto partial = torch::empty({2LL * BN_GROUPS * C}, fopts);
+  auto partial = torch::empty({2LL * G * C}, fopts);
   auto mean = torch::empty({C}, fopts);
   auto invstd = torch::empty({C}, fopts);
   launch_bn_fwd(
@@ -215,7 +225,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
       beta.has_value() ? beta->data_ptr<float>() : nullptr,
       running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
       running_var.has_value() ? running_var->data_ptr<float>() : nullptr, (float)momentum,
-      (float)eps, R, C, BN_GROUPS, dtype, stream());
+      (float)eps, R, C, G, dtype, stream());
   return {y, mean, invstd};
 }
 
@@ -225,9 +235,10 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tens
   int64_t R = x.size(0);
   int C = (int)x.size(1);
   int dtype = bn_dtype_of(x);
+  int G = bn_groups(R, C);
   auto fopts = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
   auto dx = torch::empty_like(x);
-  auto partial = torch::empty({2LL * BN_GROUPS * C}, fopts);
+  auto partial = torch::empty({2LL * G * C}, fopts);
   auto sum_dy = torch::empty({C}, fopts);
   auto sum_dy_xhat = torch::empty({C}, fopts);
   auto dgamma = torch::empty({C}, fopts);
@@ -236,7 +247,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tens
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 gamma.has_value() ? gamma->data_ptr<float>() : nullptr, sum_dy.data_ptr<float>(),
                 sum_dy_xhat.data_ptr<float>(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                R, C, BN_GROUPS, dtype, stream());
+                R, C, G, dtype, stream());
   return {dx, dgamma, dbeta};
 }
 

@@ -213,7 +213,7 @@ def test_cdna_batchnorm_matches_torch(dtype):
     g = torch.randn_like(y_ref)
     y_ref.backward(g)
     y_ours.backward(g.to(dtype))
-    assert torch.allclose(x2.grad.float(), x1.grad, atol=tol * 4), f"dx max diff {(x2.grad.float()-x1.grad).abs().max()}"
+    assert torch.allclose(x2.grad.float(), x1.grad.float(), atol=tol * 4), f"dx max diff {(x2.grad.float()-x1.grad.float()).abs().max()}"
     assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=tol * 10, rtol=1e-2)
     assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=tol * 10, rtol=1e-2)
 
