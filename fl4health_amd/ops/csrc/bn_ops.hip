@@ -59,18 +59,35 @@ __global__ __launch_bounds__(BNBLOCK) void bn_fwd_reduce_kernel(
   partial[(int64_t)(G + g) * C + c] = ss;
 }
 
-// fwd pass 2 (tiny): finalize mean/invstd, update running stats
-__global__ __launch_bounds__(BNBLOCK) void bn_fwd_finalize_kernel(
+// fwd pass 2: finalize mean/invstd, update running stats.
+// One block PER CHANNEL; 256 threads tree-reduce the G partials (G can be
+// thousands — a serial per-channel loop here was the original bottleneck).
+#define BNFIN 256
+__global__ __launch_bounds__(BNFIN) void bn_fwd_finalize_kernel(
     const float* __restrict__ partial, float* __restrict__ mean, float* __restrict__ invstd,
     float* __restrict__ running_mean, float* __restrict__ running_var, float momentum, float eps,
     int64_t R, int C, int G) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float sm_s[BNFIN];
+  __shared__ float sm_ss[BNFIN];
+  int c = blockIdx.x;
   float s = 0.0f, ss = 0.0f;
-  for (int g = 0; g < G; ++g) {
+  for (int g = threadIdx.x; g < G; g += BNFIN) {
     s += partial[(int64_t)g * C + c];
     ss += partial[(int64_t)(G + g) * C + c];
   }
+  sm_s[threadIdx.x] = s;
+  sm_ss[threadIdx.x] = ss;
+  __syncthreads();
+  for (int off = BNFIN / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sm_s[threadIdx.x] += sm_s[threadIdx.x + off];
+      sm_ss[threadIdx.x] += sm_ss[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x != 0) return;
+  s = sm_s[0];
+  ss = sm_ss[0];
   float m = s / (float)R;
   float var = fmaxf(ss / (float)R - m * m, 0.0f);
   mean[c] = m;
@@ -124,16 +141,30 @@ __global__ __launch_bounds__(BNBLOCK) void bn_bwd_reduce_kernel(
   partial[(int64_t)(G + g) * C + c] = sdyx;
 }
 
-__global__ __launch_bounds__(BNBLOCK) void bn_bwd_finalize_kernel(
+__global__ __launch_bounds__(BNFIN) void bn_bwd_finalize_kernel(
     const float* __restrict__ partial, float* __restrict__ sum_dy, float* __restrict__ sum_dy_xhat,
     float* __restrict__ dgamma, float* __restrict__ dbeta, int C, int G) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float sm_s[BNFIN];
+  __shared__ float sm_ss[BNFIN];
+  int c = blockIdx.x;
   float sdy = 0.0f, sdyx = 0.0f;
-  for (int g = 0; g < G; ++g) {
+  for (int g = threadIdx.x; g < G; g += BNFIN) {
     sdy += partial[(int64_t)g * C + c];
     sdyx += partial[(int64_t)(G + g) * C + c];
   }
+  sm_s[threadIdx.x] = sdy;
+  sm_ss[threadIdx.x] = sdyx;
+  __syncthreads();
+  for (int off = BNFIN / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sm_s[threadIdx.x] += sm_s[threadIdx.x + off];
+      sm_ss[threadIdx.x] += sm_ss[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x != 0) return;
+  sdy = sm_s[0];
+  sdyx = sm_ss[0];
   sum_dy[c] = sdy;
   sum_dy_xhat[c] = sdyx;
   if (dgamma != nullptr) dgamma[c] = sdyx;
@@ -186,9 +217,8 @@ extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mea
   } else {
     bn_fwd_reduce_kernel<float><<<grid, block, 0, s>>>((const float*)x, partial, R, C, G);
   }
-  dim3 fin_grid((C + BNBLOCK - 1) / BNBLOCK, 1, 1);
-  bn_fwd_finalize_kernel<<<fin_grid, block, 0, s>>>(partial, mean, invstd, running_mean,
-                                                    running_var, momentum, eps, R, C, G);
+  bn_fwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
+      partial, mean, invstd, running_mean, running_var, momentum, eps, R, C, G);
   // normalize: reuse G-deep row grid (bandwidth-bound)
   if (dtype == 1) {
     bn_fwd_norm_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
@@ -212,9 +242,8 @@ extern "C" void launch_bn_bwd(const void* x, const void* dy, void* dx, float* pa
     bn_bwd_reduce_kernel<float><<<grid, block, 0, s>>>((const float*)x, (const float*)dy, partial,
                                                        mean, invstd, R, C, G);
   }
-  dim3 fin_grid((C + BNBLOCK - 1) / BNBLOCK, 1, 1);
-  bn_bwd_finalize_kernel<<<fin_grid, block, 0, s>>>(partial, sum_dy, sum_dy_xhat, dgamma, dbeta, C,
-                                                    G);
+  bn_bwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
+      partial, sum_dy, sum_dy_xhat, dgamma, dbeta, C, G);
   if (dtype == 1) {
     bn_bwd_dx_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, (__hip_bfloat16*)dx, mean, invstd,
