@@ -154,3 +154,18 @@ class BasicFedAvg(Strategy):
 
     def finalize_collective(self, summed: Parameters, server_round: int, totals: dict[str, float]) -> Parameters:
         return summed
+
+
+class OpacusBasicFedAvg(BasicFedAvg):
+    """FedAvg initializing parameters from a DP-wrapped (per-sample gradient)
+    model (reference strategies/basic_fedavg.py:400: asserts GradSampleModule)."""
+
+    def __init__(self, *, model, **kwargs) -> None:
+        from fl4health_amd.common import Parameters
+        from fl4health_amd.parameter_exchange.flat import FlatParameterView
+        from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+        assert isinstance(model, GradSampleModule), "OpacusBasicFedAvg requires a GradSampleModule-wrapped model"
+        kwargs.setdefault("initial_parameters", Parameters([FlatParameterView(model.wrapped_module).flat.clone()]))
+        super().__init__(**kwargs)
+        self.model = model
