@@ -25,8 +25,7 @@ def bench_bert():
     from fl4health_amd.losses.contrastive_loss import MoonContrastiveLoss
 
     torch.manual_seed(0)
-    model = BertMoonModel(num_classes=4, small=False).cuda()
-    model = apply_lora(model, ("query", "value"), r=8)
+    model = apply_lora(BertMoonModel(num_classes=4, small=False), ("query", "value"), r=8).cuda()
     model.train()
     opt = torch.optim.AdamW([p for p in model.parameters() if p.requires_grad], lr=2e-4)
     contrastive = MoonContrastiveLoss(temperature=0.5)
