@@ -49,9 +49,11 @@ class BenchFedProxClient(FedProxClient):
     def get_model(self, config):
         model = ResNet18(num_classes=10)
         if self.device.type == "cuda":
+            from fl4health_amd.models.resnet import fuse_resnet_bn_relu
             from fl4health_amd.ops.batchnorm import convert_batchnorm_to_cdna
 
             model = convert_batchnorm_to_cdna(model.to(memory_format=torch.channels_last))
+            model = fuse_resnet_bn_relu(model)
         return model
 
     def get_data_loaders(self, config):

@@ -17,20 +17,25 @@ from fl4health_amd.ops import functional as F
 
 class _CdnaBatchNormFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2d, gamma, beta, running_mean, running_var, momentum, eps):
+    def forward(ctx, x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu):
         F._require_ext("bn_fwd_train")
-        y, mean, invstd = F._C.bn_fwd_train(x2d, gamma, beta, running_mean, running_var, momentum, eps)
-        ctx.save_for_backward(x2d, mean, invstd, gamma)
+        y, mean, invstd = F._C.bn_fwd_train(x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu)
+        ctx.save_for_backward(x2d, mean, invstd, gamma, beta)
+        ctx.fuse_relu = fuse_relu
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x2d, mean, invstd, gamma = ctx.saved_tensors
-        dx, dgamma, dbeta = F._C.bn_bwd(x2d, dy.contiguous(), mean, invstd, gamma)
-        return dx, dgamma, dbeta, None, None, None, None
+        x2d, mean, invstd, gamma, beta = ctx.saved_tensors
+        dx, dgamma, dbeta = F._C.bn_bwd(x2d, dy.contiguous(), mean, invstd, gamma, beta, ctx.fuse_relu)
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 class CdnaBatchNorm2d(nn.BatchNorm2d):
+    # when True the following ReLU is fused into the normalize kernel and the
+    # relu mask is recomputed (not stored) in backward
+    fuse_relu: bool = False
+
     def forward(self, input: torch.Tensor) -> torch.Tensor:
         use_custom = (
             input.is_cuda
@@ -52,7 +57,7 @@ class CdnaBatchNorm2d(nn.BatchNorm2d):
             self.num_batches_tracked.add_(1)
         y2d = _CdnaBatchNormFn.apply(
             x2d, self.weight.float(), self.bias.float(), self.running_mean, self.running_var,
-            float(self.momentum), float(self.eps),
+            float(self.momentum), float(self.eps), self.fuse_relu,
         )
         return y2d.view(n, h, w, c).permute(0, 3, 1, 2)
 
@@ -64,3 +69,10 @@ def convert_batchnorm_to_cdna(model: nn.Module) -> nn.Module:
         if type(m) is nn.BatchNorm2d:
             m.__class__ = CdnaBatchNorm2d
     return model
+
+
+class _FusedReluIdentity(nn.Module):
+    """Placeholder for a ReLU whose work was fused into the preceding BN."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return x

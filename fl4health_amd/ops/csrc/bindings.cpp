@@ -25,10 +25,10 @@ void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, 
                       hipStream_t);
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
 void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
-                   float*, float, float, int64_t, int, int, int, hipStream_t);
+                   float*, float, float, int64_t, int, int, int, int, hipStream_t);
 void launch_bn_bwd(const void*, const void*, void*, float*, const float*, const float*,
-                   const float*, float*, float*, float*, float*, int64_t, int, int, int,
-                   hipStream_t);
+                   const float*, const float*, float*, float*, float*, float*, int64_t, int, int,
+                   int, int, hipStream_t);
 }
 
 namespace {
@@ -207,7 +207,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
                                         c10::optional<torch::Tensor> beta,
                                         c10::optional<torch::Tensor> running_mean,
                                         c10::optional<torch::Tensor> running_var, double momentum,
-                                        double eps) {
+                                        double eps, bool fuse_relu) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2, "x must be [R, C] contiguous");
   int64_t R = x.size(0);
   int C = (int)x.size(1);
@@ -225,12 +225,13 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
       beta.has_value() ? beta->data_ptr<float>() : nullptr,
       running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
       running_var.has_value() ? running_var->data_ptr<float>() : nullptr, (float)momentum,
-      (float)eps, R, C, G, dtype, stream());
+      (float)eps, R, C, G, dtype, fuse_relu ? 1 : 0, stream());
   return {y, mean, invstd};
 }
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
-                                  torch::Tensor invstd, c10::optional<torch::Tensor> gamma) {
+                                  torch::Tensor invstd, c10::optional<torch::Tensor> gamma,
+                                  c10::optional<torch::Tensor> beta, bool fuse_relu) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous() && x.sizes() == dy.sizes());
   int64_t R = x.size(0);
   int C = (int)x.size(1);
@@ -245,9 +246,10 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tens
   auto dbeta = torch::empty({C}, fopts);
   launch_bn_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), partial.data_ptr<float>(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                gamma.has_value() ? gamma->data_ptr<float>() : nullptr, sum_dy.data_ptr<float>(),
+                gamma.has_value() ? gamma->data_ptr<float>() : nullptr,
+                beta.has_value() ? beta->data_ptr<float>() : nullptr, sum_dy.data_ptr<float>(),
                 sum_dy_xhat.data_ptr<float>(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                R, C, G, dtype, stream());
+                R, C, G, dtype, fuse_relu ? 1 : 0, stream());
   return {dx, dgamma, dbeta};
 }
 

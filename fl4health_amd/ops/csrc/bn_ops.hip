@@ -99,12 +99,12 @@ __global__ __launch_bounds__(BNFIN) void bn_fwd_finalize_kernel(
   }
 }
 
-// fwd pass 3: y = (x - mean) * invstd * gamma + beta
+// fwd pass 3: y = (x - mean) * invstd * gamma + beta, optionally fused ReLU
 template <typename T>
 __global__ __launch_bounds__(BNBLOCK) void bn_fwd_norm_kernel(
     const T* __restrict__ x, T* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ beta, int64_t R, int C) {
+    const float* __restrict__ beta, int fuse_relu, int64_t R, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float m = mean[c];
@@ -114,7 +114,9 @@ __global__ __launch_bounds__(BNBLOCK) void bn_fwd_norm_kernel(
   float scale = is * g;
   float shift = b - m * scale;
   for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
-    st<T>(y + r * C + c, fmaf(ld<T>(x + r * C + c), scale, shift));
+    float v = fmaf(ld<T>(x + r * C + c), scale, shift);
+    if (fuse_relu) v = fmaxf(v, 0.0f);
+    st<T>(y + r * C + c, v);
   }
 }
 
@@ -124,16 +126,23 @@ __global__ __launch_bounds__(BNBLOCK) void bn_fwd_norm_kernel(
 template <typename T>
 __global__ __launch_bounds__(BNBLOCK) void bn_bwd_reduce_kernel(
     const T* __restrict__ x, const T* __restrict__ dy, float* __restrict__ partial,
-    const float* __restrict__ mean, const float* __restrict__ invstd, int64_t R, int C, int G) {
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta, int fuse_relu, int64_t R,
+    int C, int G) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   int g = blockIdx.y;
   float m = mean[c];
   float is = invstd[c];
+  // with fused ReLU, dy must be masked where the pre-activation was <= 0:
+  // recompute the sign from (x_hat * gamma + beta) instead of saving y
+  float gm = gamma != nullptr ? gamma[c] : 1.0f;
+  float bt = beta != nullptr ? beta[c] : 0.0f;
   float sdy = 0.0f, sdyx = 0.0f;
   for (int64_t r = g; r < R; r += G) {
     float gy = ld<T>(dy + r * C + c);
     float xh = (ld<T>(x + r * C + c) - m) * is;
+    if (fuse_relu && fmaf(xh, gm, bt) <= 0.0f) gy = 0.0f;
     sdy += gy;
     sdyx = fmaf(gy, xh, sdyx);
   }
@@ -176,19 +185,22 @@ template <typename T>
 __global__ __launch_bounds__(BNBLOCK) void bn_bwd_dx_kernel(
     const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ gamma, const float* __restrict__ sum_dy,
-    const float* __restrict__ sum_dy_xhat, int64_t R, int C) {
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ sum_dy, const float* __restrict__ sum_dy_xhat, int fuse_relu,
+    int64_t R, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float m = mean[c];
   float is = invstd[c];
   float g = gamma != nullptr ? gamma[c] : 1.0f;
+  float bt = beta != nullptr ? beta[c] : 0.0f;
   float k = g * is;
   float mean_dy = sum_dy[c] / (float)R;
   float mean_dy_xhat = sum_dy_xhat[c] / (float)R;
   for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
     float gy = ld<T>(dy + r * C + c);
     float xh = (ld<T>(x + r * C + c) - m) * is;
+    if (fuse_relu && fmaf(xh, g, bt) <= 0.0f) gy = 0.0f;
     st<T>(dx + r * C + c, k * (gy - mean_dy - xh * mean_dy_xhat));
   }
 }
@@ -208,7 +220,7 @@ static inline void bn_dims(int C, int64_t R, int G, dim3* grid, dim3* block) {
 extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mean, float* invstd,
                               const float* gamma, const float* beta, float* running_mean,
                               float* running_var, float momentum, float eps, int64_t R, int C,
-                              int G, int dtype, hipStream_t s) {
+                              int G, int dtype, int fuse_relu, hipStream_t s) {
   dim3 grid, block;
   bn_dims(C, R, G, &grid, &block);
   if (dtype == 1) {
@@ -222,34 +234,38 @@ extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mea
   // normalize: reuse G-deep row grid (bandwidth-bound)
   if (dtype == 1) {
     bn_fwd_norm_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
-        (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, invstd, gamma, beta, R, C);
+        (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, invstd, gamma, beta, fuse_relu, R, C);
   } else {
     bn_fwd_norm_kernel<float><<<grid, block, 0, s>>>((const float*)x, (float*)y, mean, invstd,
-                                                     gamma, beta, R, C);
+                                                     gamma, beta, fuse_relu, R, C);
   }
 }
 
 extern "C" void launch_bn_bwd(const void* x, const void* dy, void* dx, float* partial,
                               const float* mean, const float* invstd, const float* gamma,
-                              float* sum_dy, float* sum_dy_xhat, float* dgamma, float* dbeta,
-                              int64_t R, int C, int G, int dtype, hipStream_t s) {
+                              const float* beta, float* sum_dy, float* sum_dy_xhat, float* dgamma,
+                              float* dbeta, int64_t R, int C, int G, int dtype, int fuse_relu,
+                              hipStream_t s) {
   dim3 grid, block;
   bn_dims(C, R, G, &grid, &block);
   if (dtype == 1) {
     bn_bwd_reduce_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
-        (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, partial, mean, invstd, R, C, G);
+        (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, partial, mean, invstd, gamma, beta,
+        fuse_relu, R, C, G);
   } else {
     bn_bwd_reduce_kernel<float><<<grid, block, 0, s>>>((const float*)x, (const float*)dy, partial,
-                                                       mean, invstd, R, C, G);
+                                                       mean, invstd, gamma, beta, fuse_relu, R, C,
+                                                       G);
   }
   bn_bwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
       partial, sum_dy, sum_dy_xhat, dgamma, dbeta, C, G);
   if (dtype == 1) {
     bn_bwd_dx_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, (__hip_bfloat16*)dx, mean, invstd,
-        gamma, sum_dy, sum_dy_xhat, R, C);
+        gamma, beta, sum_dy, sum_dy_xhat, fuse_relu, R, C);
   } else {
     bn_bwd_dx_kernel<float><<<grid, block, 0, s>>>((const float*)x, (const float*)dy, (float*)dx,
-                                                   mean, invstd, gamma, sum_dy, sum_dy_xhat, R, C);
+                                                   mean, invstd, gamma, beta, sum_dy, sum_dy_xhat,
+                                                   fuse_relu, R, C);
   }
 }

@@ -228,3 +228,31 @@ def test_cdna_batchnorm_deterministic():
     y1, m1, v1 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5)
     y2, m2, v2 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5)
     assert torch.equal(m1, m2) and torch.equal(v1, v2) and torch.equal(y1, y2)
+
+
+@requires_gpu
+def test_cdna_batchnorm_fused_relu():
+    from fl4health_amd.ops.batchnorm import CdnaBatchNorm2d
+
+    torch.manual_seed(0)
+    n, c, h, w = 8, 32, 8, 8
+    x = torch.randn(n, c, h, w, device="cuda").contiguous(memory_format=torch.channels_last)
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    ref = torch.nn.BatchNorm2d(c).cuda().train()
+    with torch.no_grad():
+        ref.weight.mul_(1.5).add_(0.1)
+        ref.bias.add_(0.2)
+    ours = CdnaBatchNorm2d(c).cuda().train()
+    ours.load_state_dict(ref.state_dict())
+    ours.fuse_relu = True
+
+    y_ref = torch.relu(ref(x1))
+    y_ours = ours(x2)
+    assert torch.allclose(y_ours, y_ref, atol=1e-4), f"{(y_ours-y_ref).abs().max()}"
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y_ours.backward(g)
+    assert torch.allclose(x2.grad, x1.grad, atol=1e-4), f"dx {(x2.grad-x1.grad).abs().max()}"
+    assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=1e-3, rtol=1e-3)
