@@ -97,3 +97,116 @@ def test_distributed_fedprox_collective_matches_simulation(tmp_path):
     for (r1, l1), (r2, l2) in zip(dist_losses, hist.losses_distributed):
         assert r1 == r2
         assert abs(l1 - l2) < 5e-4, f"round {r1}: dist {l1} vs sim {l2}"
+
+
+WORKER_SCAFFOLD = r"""
+import json, logging, torch
+logging.disable(logging.INFO)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.common import Parameters
+from fl4health_amd.clients.scaffold_client import ScaffoldClient
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.optimizers import FlatScaffoldSGD
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed
+from fl4health_amd.strategies.scaffold import Scaffold
+from tests.test_utils import TinyClient, TinyNet
+
+set_all_random_seeds(42)
+
+class Client(ScaffoldClient, TinyClient):
+    def get_optimizer(self, config):
+        return FlatScaffoldSGD(self.flat_view, lr=0.05)
+
+def strategy_factory():
+    init = Parameters([FlatParameterView(TinyNet()).flat.clone()])
+    return Scaffold(initial_parameters=init,
+                    on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 3})
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16}, strategy_factory())
+
+def client_factory(rank, world):
+    return Client(seed=rank, metrics=[Accuracy()], device="cpu")
+
+hist = run_distributed(server_factory, client_factory, num_rounds=2,
+                       strategy_factory=strategy_factory, backend="gloo")
+if hist is not None:
+    print("RESULT " + json.dumps({"losses": hist.losses_distributed}))
+"""
+
+
+def test_distributed_scaffold_packed_collective(tmp_path):
+    script = tmp_path / "worker_scaffold.py"
+    script.write_text(WORKER_SCAFFOLD)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29533",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    assert line, out.stdout[-2000:]
+    losses = json.loads(line[0][7:])["losses"]
+    assert len(losses) == 2
+    assert all(0 < l < 10 for _, l in losses)
+
+
+WORKER_PARTIAL = r"""
+import json, logging, torch
+logging.disable(logging.INFO)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed
+from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+from tests.test_utils import TinyClient
+
+set_all_random_seeds(42)
+
+def strategy_factory():
+    # half-cohort rounds: non-sampled ranks must contribute zeros correctly
+    return BasicFedAvg(fraction_fit=0.5, min_fit_clients=1, min_evaluate_clients=1,
+                       min_available_clients=1,
+                       on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 2})
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16}, strategy_factory())
+
+def client_factory(rank, world):
+    return TinyClient(seed=rank, metrics=[Accuracy()], device="cpu")
+
+hist = run_distributed(server_factory, client_factory, num_rounds=2,
+                       strategy_factory=strategy_factory, backend="gloo")
+if hist is not None:
+    print("RESULT " + json.dumps({"losses": hist.losses_distributed}))
+"""
+
+
+def test_distributed_partial_cohort(tmp_path):
+    script = tmp_path / "worker_partial.py"
+    script.write_text(WORKER_PARTIAL)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29534",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    assert line, out.stdout[-2000:]
+    losses = json.loads(line[0][7:])["losses"]
+    assert len(losses) == 2
+    assert all(0 < l < 10 for _, l in losses)
