@@ -8,6 +8,14 @@ def align_pred_and_target_shapes(preds: torch.Tensor, targets: torch.Tensor) -> 
     """Make (logits, labels) shape-compatible: one-hot vs index targets,
     trailing singleton dims, and [N] vs [N,1] mismatches."""
     if preds.shape == targets.shape:
+        # equal shapes with float one-hot-looking targets -> convert to indices
+        if (
+            preds.dim() >= 2
+            and torch.is_floating_point(targets)
+            and bool(((targets == 0) | (targets == 1)).all())
+            and bool(torch.allclose(targets.sum(dim=1), torch.ones_like(targets.sum(dim=1))))
+        ):
+            return preds, targets.argmax(dim=1)
         return preds, targets
     t = targets
     while t.dim() > 1 and t.shape[-1] == 1:

@@ -135,12 +135,20 @@ class FlatScaffoldSGD(FlatOptimizerBase):
         super().__init__(view, dict(lr=lr, weight_decay=weight_decay))
         self.lr = lr
         self.weight_decay = weight_decay
+        # persistent buffers: pointers stay fixed so a hipGraph-captured step
+        # keeps reading the CURRENT round's variates after set_variates copies
         self.c_global: torch.Tensor | None = None
         self.c_local: torch.Tensor | None = None
 
     def set_variates(self, c_global: torch.Tensor, c_local: torch.Tensor) -> None:
-        self.c_global = c_global
-        self.c_local = c_local
+        if self.c_global is None:
+            self.c_global = c_global.detach().clone().to(self.view.flat.device)
+        else:
+            self.c_global.copy_(c_global)
+        if self.c_local is None or self.c_local.data_ptr() != c_local.data_ptr():
+            # the client's own c_i tensor IS the live state: alias it so the
+            # post-round variate update is visible to the next capture-free step
+            self.c_local = c_local
 
     @torch.no_grad()
     def step(self, closure=None) -> None:  # noqa: ARG002

@@ -1,0 +1,243 @@
+"""Unit coverage for support subsystems: reporting, config, samplers,
+partitioners, data generation, feature alignment, metrics utils, LoRA,
+model bases, early stopping, accountant edge cases."""
+import json
+
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+from fl4health_amd.utils.random import set_all_random_seeds
+
+
+def test_json_reporter_schema(tmp_path):
+    from fl4health_amd.reporting.json_reporter import JsonReporter
+
+    rep = JsonReporter(run_id="testrun", output_folder=tmp_path)
+    rep.initialize(id="testrun", name="client")
+    rep.report({"fit_start": "t0"})
+    rep.report({"fit_metrics": {"acc": 0.5}}, round=1)
+    rep.report({"loss": 1.0}, round=1, step=3)
+    rep.shutdown()
+    blob = json.loads((tmp_path / "testrun.json").read_text())
+    assert blob["fit_start"] == "t0"
+    assert blob["rounds"]["1"]["fit_metrics"]["acc"] == 0.5 or blob["rounds"][1]["fit_metrics"]["acc"] == 0.5
+
+
+def test_config_validation(tmp_path):
+    from fl4health_amd.utils.config import InvalidConfigError, check_config, load_config, narrow_dict_type
+
+    with pytest.raises(InvalidConfigError):
+        check_config({"n_server_rounds": 3})
+    with pytest.raises(InvalidConfigError):
+        check_config({"n_server_rounds": -1, "batch_size": 4})
+    p = tmp_path / "c.yaml"
+    p.write_text("n_server_rounds: 3\nbatch_size: 8\nfoo: bar\n")
+    cfg = load_config(p)
+    assert narrow_dict_type(cfg, "foo", str) == "bar"
+    with pytest.raises(ValueError):
+        narrow_dict_type(cfg, "foo", int)
+
+
+def test_samplers():
+    from torch.utils.data import TensorDataset
+
+    from fl4health_amd.utils.sampler import DirichletLabelBasedSampler, MinorityLabelBasedSampler
+
+    set_all_random_seeds(0)
+    y = torch.arange(10).repeat(50)
+    ds = TensorDataset(torch.randn(500, 4), y)
+    minority = MinorityLabelBasedSampler(list(range(10)), 0.2, {0, 1})
+    sub = minority.subsample(ds)
+    sub_y = sub.tensors[1]
+    assert int((sub_y == 0).sum()) == 10 and int((sub_y == 2).sum()) == 50
+
+    dirichlet = DirichletLabelBasedSampler(list(range(10)), sample_percentage=0.5, beta=1.0, hash_key=7)
+    sub2 = dirichlet.subsample(ds)
+    assert 0 < len(sub2.tensors[1]) <= 300
+
+
+def test_dirichlet_partitioner_sizes():
+    from fl4health_amd.datasets.partitioners import DirichletLabelPartitioner
+
+    labels = torch.randint(0, 10, (1000,))
+    parts = DirichletLabelPartitioner(4, beta=0.5, min_size=5, seed=3).partition_indices(labels)
+    assert len(parts) == 4
+    assert sum(len(p) for p in parts) == 1000
+    all_idx = np.concatenate(parts)
+    assert len(np.unique(all_idx)) == 1000
+
+
+def test_synthetic_fedprox_generator():
+    from fl4health_amd.utils.data_generation import SyntheticFedProxDataset, SyntheticIidFedProxDataset
+
+    gen = SyntheticFedProxDataset(3, alpha=1.0, beta=1.0, samples_per_client=50, seed=1)
+    dsets = gen.generate()
+    assert len(dsets) == 3
+    x, y = dsets[0].tensors
+    assert x.shape == (50, 60) and y.max() < 10
+
+    iid = SyntheticIidFedProxDataset(2, samples_per_client=20, seed=1).generate()
+    assert len(iid) == 2
+
+
+def test_feature_alignment_cross_schema():
+    import pandas as pd
+
+    from fl4health_amd.feature_alignment.tab_features_info_encoder import TabularFeaturesInfoEncoder
+    from fl4health_amd.feature_alignment.tab_features_preprocessor import TabularFeaturesPreprocessor
+
+    df_a = pd.DataFrame({"num": [1.0, 2.0, 3.0], "cat": ["x", "y", "x"], "label": [0, 1, 0]})
+    df_b = pd.DataFrame({"num": [5.0, 6.0], "label": [1, 0]})  # missing 'cat'
+    enc = TabularFeaturesInfoEncoder.encoder_from_dataframe(df_a, None, "label")
+    enc2 = TabularFeaturesInfoEncoder.from_json(enc.to_json())
+    pre = TabularFeaturesPreprocessor(enc2)
+    xa, ya = pre.preprocess(df_a)
+    xb, yb = pre.preprocess(df_b)
+    assert xa.shape[1] == xb.shape[1] == enc.input_dimension()
+    assert (xb[:, -2:] == 0).all()  # missing categorical -> zero block
+
+
+def test_metrics_utils_alignment():
+    from fl4health_amd.metrics.utils import align_pred_and_target_shapes, map_label_index_tensor_to_one_hot
+
+    preds = torch.randn(4, 3)
+    targets = torch.tensor([[0.0, 1.0, 0.0]] * 4)
+    p, t = align_pred_and_target_shapes(preds, targets)
+    assert t.shape == (4,)
+    one_hot = map_label_index_tensor_to_one_hot(torch.tensor([0, 2]), (2, 3))
+    assert one_hot.shape == (2, 3)
+
+
+def test_lora_merge_equivalence():
+    from fl4health_amd.models.lora import LoraLinear
+
+    set_all_random_seeds(0)
+    base = nn.Linear(8, 6)
+    lora = LoraLinear(base, r=2, alpha=4)
+    with torch.no_grad():
+        lora.lora_B.normal_()
+    x = torch.randn(5, 8)
+    before = lora(x)
+    lora.merge_weights()
+    after = lora.base(x)
+    assert torch.allclose(before, after, atol=1e-5)
+    assert not lora.base.weight.requires_grad
+    assert lora.lora_A.requires_grad
+
+
+def test_ensemble_vote_mode():
+    from fl4health_amd.model_bases.ensemble_base import EnsembleAggregationMode, EnsembleModel
+
+    m = EnsembleModel({"a": nn.Linear(4, 3), "b": nn.Linear(4, 3)}, EnsembleAggregationMode.VOTE)
+    out = m(torch.randn(6, 4))
+    assert out["ensemble-pred"].shape == (6, 3)
+    assert torch.allclose(out["ensemble-pred"].sum(dim=1), torch.ones(6))
+
+
+def test_pca_module_roundtrip():
+    from fl4health_amd.model_bases.pca import PcaModule
+
+    set_all_random_seeds(0)
+    x = torch.randn(50, 10) @ torch.randn(10, 10)
+    pca = PcaModule()
+    pcs, svs = pca(x)
+    assert pca.compute_cumulative_explained_variance(10) > 0.99
+    proj = pca.project_lower_dim(x, 5, center_data=True)
+    assert proj.shape == (50, 5)
+    err_full = pca.compute_reconstruction_error(x, 10)
+    err_small = pca.compute_reconstruction_error(x, 2)
+    assert err_small >= err_full
+
+
+def test_vae_loss_and_model():
+    from fl4health_amd.model_bases.autoencoders_base import VariationalAe
+    from fl4health_amd.preprocessing.autoencoders import VaeLoss
+
+    class Enc(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.mu = nn.Linear(6, 2)
+            self.logvar = nn.Linear(6, 2)
+
+        def forward(self, x):
+            return self.mu(x), self.logvar(x)
+
+    vae = VariationalAe(Enc(), nn.Linear(2, 6))
+    x = torch.randn(4, 6)
+    out = vae(x)
+    assert out.shape == (4, 6 + 2 + 2)
+    loss = VaeLoss(latent_dim=2)(out, x)
+    assert torch.isfinite(loss)
+
+
+def test_gpfl_components():
+    from fl4health_amd.model_bases.gpfl_base import CoV, Gce
+
+    gce = Gce(8, 4)
+    f = torch.randn(6, 8)
+    y = torch.randint(0, 4, (6,))
+    loss = gce(f, y)
+    assert torch.isfinite(loss)
+    emb = gce.lookup(y)
+    assert emb.shape == (6, 8)
+    cov = CoV(8)
+    out = cov(f, torch.randn(6, 8))
+    assert out.shape == (6, 8) and (out >= 0).all()
+
+
+def test_early_stopper_restores_best(tmp_path):
+    from fl4health_amd.utils.early_stopper import EarlyStopper
+    from fl4health_amd.metrics.metrics import Accuracy
+    from tests.test_utils import TinyClient
+
+    set_all_random_seeds(0)
+    client = TinyClient(seed=0, metrics=[Accuracy()], device="cpu")
+    client.setup_client({"batch_size": 8})
+    stopper = EarlyStopper(client, patience=1, interval_steps=1, snapshot_dir=tmp_path)
+    assert stopper.should_stop(1) is False  # first eval snapshots
+    # worsen the model drastically -> patience hits
+    with torch.no_grad():
+        for p in client.model.parameters():
+            p.mul_(100.0)
+    stopped = stopper.should_stop(2)
+    assert stopped is True
+
+
+def test_accountant_poisson_vs_fixed():
+    from fl4health_amd.privacy.fl_accountants import (
+        FlClientLevelAccountantFixedSamplingNoReplacement,
+        FlClientLevelAccountantPoissonSampling,
+    )
+
+    p = FlClientLevelAccountantPoissonSampling(0.1, 1.0)
+    f = FlClientLevelAccountantFixedSamplingNoReplacement(100, 10, 1.0)
+    ep = p.get_epsilon(100, 1e-5)
+    ef = f.get_epsilon(100, 1e-5)
+    assert 0 < ep < 100 and 0 < ef < 100
+
+
+def test_warmed_up_module(tmp_path):
+    from fl4health_amd.preprocessing.warmed_up_module import WarmedUpModule
+
+    src = nn.Sequential(nn.Linear(4, 4))
+    dst = nn.Sequential(nn.Linear(4, 4))
+    warm = WarmedUpModule(pretrained_model=src)
+    warm.load_from_pretrained(dst)
+    assert torch.allclose(src[0].weight, dst[0].weight)
+
+
+def test_device_tensor_loader_epochs():
+    from fl4health_amd.datasets.loaders import DeviceTensorLoader
+
+    x = torch.randn(10, 3)
+    y = torch.arange(10)
+    loader = DeviceTensorLoader(x, y, batch_size=4, device="cpu", shuffle=True, drop_last=True, seed=0)
+    assert len(loader) == 2
+    seen = [yy for _, yy in loader]
+    assert sum(t.numel() for t in seen) == 8
+    # second epoch reshuffles
+    e1 = torch.cat([yy for _, yy in loader])
+    e2 = torch.cat([yy for _, yy in loader])
+    assert not torch.equal(e1, e2)
