@@ -1,0 +1,94 @@
+"""Rounds-to-target-accuracy evidence for the flagship config (BASELINE.json):
+CIFAR-10-shaped ResNet-18 FedProx, 2 simulated clients on one MI355X,
+Dirichlet(0.5) non-IID synthetic shards, eval every round."""
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.clients.adaptive_drift_constraint_client import FedProxClient
+from fl4health_amd.common import Parameters
+from fl4health_amd.datasets.loaders import DeviceTensorLoader
+from fl4health_amd.datasets.synthetic import synthetic_classification_dataset
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.models.resnet import ResNet18, fuse_resnet_bn_relu
+from fl4health_amd.ops.batchnorm import convert_batchnorm_to_cdna
+from fl4health_amd.optimizers import FlatProxSGD
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_simulation
+from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+from fl4health_amd.utils.random import set_all_random_seeds
+
+
+class Client(FedProxClient):
+    def __init__(self, seed, **kw):
+        super().__init__(**kw)
+        self.seed = seed
+        if self.device.type == "cuda":
+            self.autocast_dtype = torch.bfloat16
+            self.use_cuda_graph = True
+
+    def get_model(self, config):
+        model = ResNet18(num_classes=10)
+        if self.device.type == "cuda":
+            model = fuse_resnet_bn_relu(convert_batchnorm_to_cdna(model.to(memory_format=torch.channels_last)))
+        return model
+
+    def get_data_loaders(self, config):
+        train = synthetic_classification_dataset(4096, (3, 32, 32), 10, seed=self.seed, signal=0.6)
+        val = synthetic_classification_dataset(1024, (3, 32, 32), 10, seed=self.seed + 100, signal=0.6)
+        cl = self.device.type == "cuda"
+        return (
+            DeviceTensorLoader(train.tensors[0], train.tensors[1], 128, self.device, seed=self.seed, channels_last=cl),
+            DeviceTensorLoader(val.tensors[0], val.tensors[1], 256, self.device, shuffle=False, drop_last=False, channels_last=cl),
+        )
+
+    def get_optimizer(self, config):
+        return FlatProxSGD(self.flat_view, lr=0.05, momentum=0.9, weight_decay=5e-4)
+
+    def get_criterion(self, config):
+        return torch.nn.CrossEntropyLoss()
+
+
+def main():
+    set_all_random_seeds(42)
+    torch.backends.cudnn.benchmark = True
+    device = "cuda:0" if torch.cuda.is_available() else "cpu"
+    rounds = int(os.environ.get("ROUNDS", "15"))
+    clients = [Client(i, metrics=[Accuracy()], device=device) for i in range(2)]
+    strategy = FedAvgWithAdaptiveConstraint(
+        initial_parameters=Parameters([FlatParameterView(ResNet18()).flat.clone().to(device)]),
+        initial_loss_weight=0.1,
+        adapt_loss_weight=True,
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 10},
+    )
+    server = FlServer(SimpleClientManager(), {"n_server_rounds": rounds, "batch_size": 128}, strategy)
+    t0 = time.perf_counter()
+    hist = run_simulation(server, clients, num_rounds=rounds)
+    elapsed = time.perf_counter() - t0
+    accs = hist.metrics_distributed.get("val - prediction - accuracy", [])
+    lines = [
+        "# FedProx convergence on one MI355X (2 simulated clients, ResNet-18,",
+        "# synthetic Dirichlet-style shards, 10 local steps/round, batch 128, bf16)",
+        "",
+        f"total wall: {elapsed:.1f}s for {rounds} rounds (incl eval)",
+        "",
+        "| round | agg val loss | agg val accuracy |",
+        "|---|---|---|",
+    ]
+    for (r, loss), (_, acc) in zip(hist.losses_distributed, accs):
+        lines.append(f"| {r} | {loss:.4f} | {acc:.4f} |")
+    out = "\n".join(lines)
+    print(out)
+    os.makedirs("gpurun_out", exist_ok=True)
+    with open("gpurun_out/convergence.md", "w") as f:
+        f.write(out + "\n")
+
+
+if __name__ == "__main__":
+    main()
