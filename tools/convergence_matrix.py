@@ -22,7 +22,7 @@ from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithA
 from fl4health_amd.utils.random import set_all_random_seeds
 
 
-def run(opt_kind, lr, momentum, use_graph, use_cdna_bn, rounds=5):
+def run(opt_kind, lr, momentum, use_graph, use_cdna_bn, rounds=5, fuse=True):
     set_all_random_seeds(42)
     device = "cuda:0"
 
@@ -36,7 +36,9 @@ def run(opt_kind, lr, momentum, use_graph, use_cdna_bn, rounds=5):
         def get_model(self, config):
             m = ResNet18(num_classes=10).to(memory_format=torch.channels_last)
             if use_cdna_bn:
-                m = fuse_resnet_bn_relu(convert_batchnorm_to_cdna(m))
+                m = convert_batchnorm_to_cdna(m)
+                if fuse:
+                    m = fuse_resnet_bn_relu(m)
             return m
 
         def get_data_loaders(self, config):
@@ -70,8 +72,6 @@ def run(opt_kind, lr, momentum, use_graph, use_cdna_bn, rounds=5):
 
 if __name__ == "__main__":
     torch.backends.cudnn.benchmark = True
-    run("torch", 0.05, 0.9, False, False)
-    run("flat", 0.05, 0.9, False, False)
-    run("flat", 0.05, 0.9, True, False)
-    run("flat", 0.05, 0.9, True, True)
-    run("flat", 0.01, 0.0, True, True)
+    run("flat", 0.05, 0.9, False, True, fuse=False)   # BN alone, no graph, no fusion
+    run("flat", 0.05, 0.9, False, True, fuse=True)    # BN+fusion, no graph
+    run("flat", 0.05, 0.9, True, True, fuse=False)    # BN+graph, no fusion
