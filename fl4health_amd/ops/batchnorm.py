@@ -49,7 +49,10 @@ class CdnaBatchNorm2d(nn.BatchNorm2d):
             and F.HAS_EXT
         )
         if not use_custom:
-            return super().forward(input)
+            out = super().forward(input)
+            # fuse_relu moves the following ReLU INTO this module: the
+            # fallback (eval / CPU / non-channels-last) path must apply it too
+            return torch.relu(out) if self.fuse_relu else out
         n, c, h, w = input.shape
         # channels-last memory IS [N*H*W, C] row-major
         x2d = input.permute(0, 2, 3, 1).reshape(n * h * w, c)

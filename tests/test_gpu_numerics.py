@@ -256,3 +256,19 @@ def test_cdna_batchnorm_fused_relu():
     assert torch.allclose(x2.grad, x1.grad, atol=1e-4), f"dx {(x2.grad-x1.grad).abs().max()}"
     assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(ours.bias.grad, ref.bias.grad, atol=1e-3, rtol=1e-3)
+
+
+@requires_gpu
+def test_cdna_batchnorm_fused_relu_eval_path():
+    from fl4health_amd.ops.batchnorm import CdnaBatchNorm2d
+
+    torch.manual_seed(0)
+    bn = CdnaBatchNorm2d(8).cuda()
+    bn.fuse_relu = True
+    x = torch.randn(4, 8, 4, 4, device="cuda").contiguous(memory_format=torch.channels_last)
+    bn.train()
+    y_train = bn(x)
+    assert (y_train >= 0).all()
+    bn.eval()
+    y_eval = bn(x)
+    assert (y_eval >= 0).all(), "eval fallback must apply the fused ReLU"
