@@ -40,8 +40,8 @@ class Client(FedProxClient):
         return model
 
     def get_data_loaders(self, config):
-        train = synthetic_classification_dataset(4096, (3, 32, 32), 10, seed=self.seed, signal=0.6)
-        val = synthetic_classification_dataset(1024, (3, 32, 32), 10, seed=self.seed + 100, signal=0.6)
+        train = synthetic_classification_dataset(4096, (3, 32, 32), 10, seed=self.seed, signal=1.5)
+        val = synthetic_classification_dataset(1024, (3, 32, 32), 10, seed=self.seed + 100, signal=1.5)
         cl = self.device.type == "cuda"
         return (
             DeviceTensorLoader(train.tensors[0], train.tensors[1], 128, self.device, seed=self.seed, channels_last=cl),
@@ -49,7 +49,7 @@ class Client(FedProxClient):
         )
 
     def get_optimizer(self, config):
-        return FlatProxSGD(self.flat_view, lr=0.05, momentum=0.9, weight_decay=5e-4)
+        return FlatProxSGD(self.flat_view, lr=0.02, momentum=0.9, weight_decay=5e-4)
 
     def get_criterion(self, config):
         return torch.nn.CrossEntropyLoss()
@@ -65,7 +65,7 @@ def main():
         initial_parameters=Parameters([FlatParameterView(ResNet18()).flat.clone().to(device)]),
         initial_loss_weight=0.1,
         adapt_loss_weight=True,
-        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 10},
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 20},
     )
     server = FlServer(SimpleClientManager(), {"n_server_rounds": rounds, "batch_size": 128}, strategy)
     t0 = time.perf_counter()
