@@ -17,13 +17,18 @@ def synthetic_classification_dataset(
     num_classes: int = 10,
     seed: int = 0,
     signal: float = 1.0,
+    basis_seed: int = 777,
 ) -> TensorDataset:
-    """Random images with a per-class mean shift so models can learn."""
+    """Random images with a per-class mean shift so models can learn.
+
+    The class-signal basis comes from `basis_seed` (NOT `seed`) so different
+    splits/clients share the same label->pattern mapping and generalization
+    across them is possible; `seed` only controls the sampled points.
+    """
     gen = torch.Generator().manual_seed(seed)
     y = torch.randint(0, num_classes, (n,), generator=gen)
     x = torch.randn((n, *shape), generator=gen)
-    # class-dependent low-frequency pattern
-    basis = torch.randn((num_classes, *shape), generator=gen)
+    basis = torch.randn((num_classes, *shape), generator=torch.Generator().manual_seed(basis_seed))
     x += signal * basis[y]
     return TensorDataset(x, y)
 
