@@ -40,8 +40,8 @@ class Client(FedProxClient):
         return model
 
     def get_data_loaders(self, config):
-        train = synthetic_classification_dataset(4096, (3, 32, 32), 10, seed=self.seed, signal=1.5)
-        val = synthetic_classification_dataset(1024, (3, 32, 32), 10, seed=self.seed + 100, signal=1.5)
+        train = synthetic_classification_dataset(4096, (3, 32, 32), 10, seed=self.seed, signal=0.25)
+        val = synthetic_classification_dataset(1024, (3, 32, 32), 10, seed=self.seed + 100, signal=0.25)
         cl = self.device.type == "cuda"
         return (
             DeviceTensorLoader(train.tensors[0], train.tensors[1], 128, self.device, seed=self.seed, channels_last=cl),
