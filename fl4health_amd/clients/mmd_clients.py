@@ -98,11 +98,37 @@ class DittoMkMmdClient(_MmdMixin, DittoClient):
         super().update_after_step(step, current_round)
 
     def train_step(self, input, target):
-        losses, preds = super().train_step(input, target)
-        if self.mmd_loss_weight != 0 and self.mmd_losses:
-            penalty, per_layer = self._mmd_penalty(input)
-            losses.additional_losses.update(per_layer)
-            losses.additional_losses["total_mmd_loss"] = penalty.detach()
+        """Ditto tandem step with the MMD penalty inside the LOCAL backward."""
+        import torch as _torch
+
+        from fl4health_amd.optimizers import FlatProxSGD
+        from fl4health_amd.utils.losses import TrainingLosses
+
+        if not (self.mmd_loss_weight != 0 and self.mmd_losses):
+            return super().train_step(input, target)
+        self.set_optimizer_zero_grad()
+        preds, _ = self.predict(input)
+        target = self.transform_target(target)
+        global_loss = self.criterion(preds["global"], target)
+        local_loss = self.criterion(preds["prediction"], target)
+        penalty, per_layer = self._mmd_penalty(input)
+        total_local = local_loss + self.mmd_loss_weight * penalty
+        global_loss.backward()
+        total_local.backward()
+        self.optimizers["global"].step()
+        self.optimizers["local"].step()
+        local_opt = self.optimizers.get("local")
+        drift = local_opt.drift_loss() if isinstance(local_opt, FlatProxSGD) else _torch.zeros(())
+        losses = TrainingLosses(
+            backward={"backward": (total_local + drift).detach()},
+            additional_losses={
+                "global_loss": global_loss.detach(),
+                "local_loss": local_loss.detach(),
+                "total_mmd_loss": penalty.detach(),
+                **per_layer,
+            },
+        )
+        self._vanilla_loss_for_packing = float(global_loss.detach())
         return losses, preds
 
 
@@ -128,8 +154,19 @@ class MrMtlMkMmdClient(_MmdMixin, MrMtlClient):
 
     def compute_training_loss(self, preds, features, target) -> TrainingLosses:
         losses = super().compute_training_loss(preds, features, target)
-        # MMD penalty added to the backward loss through additional bookkeeping
+        if self.mmd_loss_weight != 0 and self.mmd_losses and self._last_input is not None:
+            penalty, per_layer = self._mmd_penalty(self._last_input)
+            # MR-MTL trains a single model: fold the penalty into the backward loss
+            losses.backward["backward"] = losses.backward["backward"] + self.mmd_loss_weight * penalty
+            losses.additional_losses.update(per_layer)
+            losses.additional_losses["total_mmd_loss"] = penalty.detach()
         return losses
+
+    def predict(self, input):
+        self._last_input = input
+        return super().predict(input)
+
+    _last_input = None
 
 
 class DittoDeepMmdClient(DittoMkMmdClient):

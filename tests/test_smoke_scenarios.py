@@ -478,3 +478,33 @@ def test_smoke_bert_moon_lora():
     # only adapter + head weights crossed the boundary
     p = clients[0].get_parameters({"current_server_round": 1})
     assert all(("lora" in n) or n.startswith("head.") for n in p.meta["layer_names"])
+
+
+def test_smoke_mkmmd_clients():
+    from fl4health_amd.clients.mmd_clients import DittoMkMmdClient
+    from fl4health_amd.optimizers import FlatProxSGD
+    from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+
+    set_all_random_seeds(42)
+
+    class Client(DittoMkMmdClient, TinyClient):
+        def get_optimizer(self, config):
+            return {"local": FlatProxSGD(self.flat_view, lr=0.05), "global": None}
+
+        def setup_client(self, config):
+            super().setup_client(config)
+            self.optimizers["global"] = FlatProxSGD(self.global_flat_view, lr=0.05)
+
+    clients = [
+        Client(
+            seed=i, n_train=N_TRAIN, metrics=[Accuracy()], device="cpu",
+            flatten_feature_extraction_layers={"conv": True},
+            mkmmd_loss_weight=1.0, beta_global_update_interval=2,
+        )
+        for i in range(2)
+    ]
+    strategy = FedAvgWithAdaptiveConstraint(
+        initial_parameters=_init_params(TinyNet), initial_loss_weight=0.5, on_fit_config_fn=_fit_cfg
+    )
+    hist = _run(FlServer(SimpleClientManager(), CFG, strategy), clients)
+    assert hist is not None
