@@ -42,6 +42,9 @@ class NumpyClippingClient(BasicClient):
         return clipped_delta, float(bit[0].item()) if bit is not None else 1.0
 
     def get_parameters(self, config: Config) -> Parameters:
+        if not self.initialized:
+            # round-0 initialization handshake: full weights, no aux
+            return self.setup_client_and_return_all_model_parameters(config)
         assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
         clipped_delta, bit = self.compute_weight_update_and_clip()
         # client-level DP sends clipped weight DELTAS, not weights

@@ -68,6 +68,9 @@ class DittoClient(BasicClient):
                 local_opt.set_penalty_weight(self.lam)
 
     def get_parameters(self, config: Config) -> Parameters:
+        if not self.initialized:
+            # round-0 initialization handshake: full weights, no aux
+            return self.setup_client_and_return_all_model_parameters(config)
         assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
         model_params = self.parameter_exchanger.push_parameters(self.global_model, config=config)
         return self.parameter_exchanger.pack_parameters(model_params, self._vanilla_loss_for_packing)

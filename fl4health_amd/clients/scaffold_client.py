@@ -78,6 +78,9 @@ class ScaffoldClient(BasicClient):
         super().update_after_train(local_steps, loss_dict, config)
 
     def get_parameters(self, config: Config) -> Parameters:
+        if not self.initialized:
+            # round-0 initialization handshake: full weights, no aux
+            return self.setup_client_and_return_all_model_parameters(config)
         assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
         model_params = self.parameter_exchanger.push_parameters(self.model, config=config)
         assert self.client_control_variates_updates is not None, "fit must run before get_parameters"
@@ -88,6 +91,18 @@ class ScaffoldClient(BasicClient):
         return self.parameter_exchanger.pack_parameters(model_params, full)
 
 
-class DPScaffoldClient(ScaffoldClient):
-    """SCAFFOLD + instance-level DP-SGD (reference scaffold_client.py:297):
-    combined in clients/instance_level_dp_client.py integration."""
+from fl4health_amd.clients.instance_level_dp_client import InstanceLevelDpClient  # noqa: E402
+
+
+class DPScaffoldClient(InstanceLevelDpClient, ScaffoldClient):
+    """SCAFFOLD + instance-level DP-SGD (reference scaffold_client.py:297,
+    multiple inheritance like the reference): per-sample clipped + noised
+    gradients land in the flat grad buffer (DP engine writes INTO the bound
+    .grad views), then the fused variate-corrected SGD kernel applies
+    p -= lr*(g_dp + c - c_i)."""
+
+    def setup_client(self, config) -> None:
+        super().setup_client(config)
+        assert isinstance(self.optimizers["global"], FlatScaffoldSGD), (
+            "DPScaffoldClient requires a FlatScaffoldSGD optimizer"
+        )

@@ -508,3 +508,34 @@ def test_smoke_mkmmd_clients():
     )
     hist = _run(FlServer(SimpleClientManager(), CFG, strategy), clients)
     assert hist is not None
+
+
+def test_smoke_dp_scaffold():
+    from fl4health_amd.clients.scaffold_client import DPScaffoldClient
+    from fl4health_amd.optimizers import FlatScaffoldSGD
+    from fl4health_amd.servers.scaffold_server import DPScaffoldServer
+    from fl4health_amd.strategies.scaffold import Scaffold
+
+    set_all_random_seeds(42)
+
+    class Client(DPScaffoldClient, TinyClient):
+        def get_optimizer(self, config):
+            return FlatScaffoldSGD(self.flat_view, lr=0.05)
+
+    clients = [
+        Client(seed=i, n_train=N_TRAIN, metrics=[Accuracy()], device="cpu",
+               clipping_bound=5.0, noise_multiplier=0.1)
+        for i in range(2)
+    ]
+    # initial params must match the DP-converted architecture (BN -> GroupNorm)
+    from fl4health_amd.privacy.grad_sample import convert_batchnorm_modules
+
+    strategy = Scaffold(
+        initial_parameters=_init_params(lambda: convert_batchnorm_modules(TinyNet())),
+        on_fit_config_fn=_fit_cfg,
+    )
+    server = DPScaffoldServer(
+        SimpleClientManager(), CFG, strategy, noise_multiplier=0.1, local_steps=STEPS
+    )
+    hist = _run(server, clients)
+    assert float(strategy.server_control_variates.abs().sum()) > 0
