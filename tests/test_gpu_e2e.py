@@ -103,3 +103,74 @@ def test_dp_sgd_engine_on_gpu():
     torch.cuda.synchronize()
     for p, pr in zip(model.parameters(), ref_model.parameters()):
         assert torch.allclose(p, pr, atol=1e-4), f"max diff {(p - pr).abs().max()}"
+
+
+@requires_gpu
+def test_fedpm_masked_training_on_gpu():
+    from fl4health_amd.model_bases.masked_layers import convert_to_masked_model
+    from fl4health_amd.models.cnn import SmallCnn
+
+    torch.manual_seed(0)
+    model = convert_to_masked_model(SmallCnn()).cuda()
+    opt = torch.optim.Adam([p for p in model.parameters() if p.requires_grad], lr=0.01)
+    x = torch.randn(16, 3, 32, 32, device="cuda")
+    y = torch.randint(0, 10, (16,), device="cuda")
+    scores_before = model.conv1.weight_scores.detach().clone()
+    for _ in range(3):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+    torch.cuda.synchronize()
+    assert not torch.allclose(model.conv1.weight_scores.detach(), scores_before)
+    assert model.conv1.weight.grad is None  # weights frozen
+
+
+@requires_gpu
+def test_dp_scaffold_round_on_gpu():
+    from fl4health_amd.client_managers.base import SimpleClientManager
+    from fl4health_amd.clients.scaffold_client import DPScaffoldClient
+    from fl4health_amd.common import Parameters
+    from fl4health_amd.datasets.synthetic import synthetic_cifar_loaders
+    from fl4health_amd.metrics.metrics import Accuracy
+    from fl4health_amd.models.cnn import SmallCnn
+    from fl4health_amd.optimizers import FlatScaffoldSGD
+    from fl4health_amd.parameter_exchange.flat import FlatParameterView
+    from fl4health_amd.privacy.grad_sample import convert_batchnorm_modules
+    from fl4health_amd.servers.scaffold_server import DPScaffoldServer
+    from fl4health_amd.simulation import run_simulation
+    from fl4health_amd.strategies.scaffold import Scaffold
+    from fl4health_amd.utils.random import set_all_random_seeds
+
+    set_all_random_seeds(42)
+
+    class Client(DPScaffoldClient):
+        def __init__(self, seed, **kw):
+            super().__init__(**kw)
+            self.seed = seed
+
+        def get_model(self, config):
+            return SmallCnn()
+
+        def get_data_loaders(self, config):
+            return synthetic_cifar_loaders(n_train=128, n_val=64, batch_size=16, seed=self.seed)
+
+        def get_optimizer(self, config):
+            return FlatScaffoldSGD(self.flat_view, lr=0.05)
+
+        def get_criterion(self, config):
+            return torch.nn.CrossEntropyLoss()
+
+    clients = [Client(i, metrics=[Accuracy()], device="cuda:0", clipping_bound=2.0, noise_multiplier=0.3) for i in range(2)]
+    init = Parameters([FlatParameterView(convert_batchnorm_modules(SmallCnn())).flat.clone().cuda()])
+    strategy = Scaffold(
+        initial_parameters=init,
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 2},
+    )
+    server = DPScaffoldServer(
+        SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16}, strategy,
+        noise_multiplier=0.3, local_steps=2,
+    )
+    hist = run_simulation(server, clients, num_rounds=2)
+    assert len(hist.losses_distributed) == 2
+    assert float(strategy.server_control_variates.abs().sum()) > 0
