@@ -12,7 +12,7 @@ GOLDEN_DIR = Path(__file__).resolve().parent / "golden"
 DEFAULT_TOL = 5e-4
 
 
-@pytest.mark.parametrize("name", ["fedavg", "fedprox", "scaffold"])
+@pytest.mark.parametrize("name", ["fedavg", "fedprox", "scaffold", "ditto", "apfl", "moon"])
 def test_golden_metrics(name):
     with open(GOLDEN_DIR / f"{name}_golden.json") as f:
         golden = json.load(f)

@@ -55,6 +55,55 @@ def run_scenario(name):
             initial_parameters=Parameters([FlatParameterView(TinyNet()).flat.clone()]),
             on_fit_config_fn=fit_cfg,
         )
+    elif name == "ditto":
+        from fl4health_amd.clients.ditto_client import DittoClient
+        from fl4health_amd.optimizers import FlatProxSGD
+        from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+
+        class C(DittoClient, TinyClient):
+            def get_optimizer(self, config):
+                return {"local": FlatProxSGD(self.flat_view, lr=0.05), "global": None}
+
+            def setup_client(self, config):
+                super().setup_client(config)
+                self.optimizers["global"] = FlatProxSGD(self.global_flat_view, lr=0.05)
+
+        clients = [C(seed=i, n_train=128, metrics=[Accuracy()], device="cpu") for i in range(2)]
+        strategy = FedAvgWithAdaptiveConstraint(
+            initial_parameters=Parameters([FlatParameterView(TinyNet()).flat.clone()]),
+            initial_loss_weight=0.5, on_fit_config_fn=fit_cfg,
+        )
+    elif name == "apfl":
+        from fl4health_amd.clients.apfl_client import ApflClient
+        from fl4health_amd.model_bases.apfl_base import ApflModule
+        from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+
+        class C(ApflClient, TinyClient):
+            def get_model(self, config):
+                return ApflModule(TinyNet(), adaptive_alpha=True)
+
+            def get_optimizer(self, config):
+                return {
+                    "global": torch.optim.SGD(self.model.global_model.parameters(), lr=0.05),
+                    "local": torch.optim.SGD(self.model.local_model.parameters(), lr=0.05),
+                }
+
+        clients = [C(seed=i, n_train=128, metrics=[Accuracy()], device="cpu") for i in range(2)]
+        strategy = BasicFedAvg(on_fit_config_fn=fit_cfg)
+    elif name == "moon":
+        import torch.nn as nn
+
+        from fl4health_amd.clients.moon_client import MoonClient
+        from fl4health_amd.model_bases.moon_base import MoonModel
+        from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+
+        class C(MoonClient, TinyClient):
+            def get_model(self, config):
+                base = nn.Sequential(nn.Conv2d(3, 4, 3, padding=1), nn.ReLU(), nn.Flatten())
+                return MoonModel(base, nn.Linear(4 * 32 * 32, 10))
+
+        clients = [C(seed=i, n_train=128, metrics=[Accuracy()], device="cpu") for i in range(2)]
+        strategy = BasicFedAvg(on_fit_config_fn=fit_cfg)
     else:
         raise ValueError(name)
 
@@ -68,7 +117,7 @@ def run_scenario(name):
 
 def main():
     GOLDEN_DIR.mkdir(parents=True, exist_ok=True)
-    for name in ("fedavg", "fedprox", "scaffold"):
+    for name in ("fedavg", "fedprox", "scaffold", "ditto", "apfl", "moon"):
         result = run_scenario(name)
         path = GOLDEN_DIR / f"{name}_golden.json"
         with open(path, "w") as f:
