@@ -57,15 +57,19 @@ def bench_bert():
     log(f"BERT-base + LoRA + MOON (batch {batch}, seq {seq}, bf16): {ms:.1f} ms/step = {toks/1e3:.1f}k tokens/s/GPU")
 
 
-def bench_unet(patch=128, base=32, levels=5, batch=2):
+def bench_unet(patch=128, base=32, levels=5, batch=2, channels_last=False):
     from fl4health_amd.models.unet3d import DeepSupervisionLoss, UNet3D
 
     torch.manual_seed(0)
     model = UNet3D(1, 3, base_channels=base, num_levels=levels, deep_supervision=True).cuda()
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last_3d)
     model.train()
     opt = torch.optim.SGD(model.parameters(), lr=1e-2, momentum=0.99, nesterov=True)
     crit = DeepSupervisionLoss(3)
     x = torch.randn(batch, 1, patch, patch, patch, device="cuda")
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last_3d)
     y = torch.randint(0, 3, (batch, patch, patch, patch), device="cuda")
 
     def step():
@@ -87,14 +91,18 @@ def bench_unet(patch=128, base=32, levels=5, batch=2):
     torch.cuda.synchronize()
     ms = (time.perf_counter() - t0) / n * 1e3
     vox = batch * patch**3 / (ms / 1e3)
-    log(f"3D U-Net {patch}^3 (base {base}, {levels} levels, batch {batch}, bf16, deep supervision): "
+    log(f"3D U-Net {patch}^3 (base {base}, {levels} levels, batch {batch}, bf16, ds, cl={channels_last}): "
         f"{ms:.1f} ms/step = {vox/1e6:.1f}M voxels/s/GPU; peak mem {torch.cuda.max_memory_allocated()/2**30:.1f} GiB")
 
 
 if __name__ == "__main__":
-    bench_bert()
+    import os
+    if os.environ.get("SKIP_BERT") != "1":
+        bench_bert()
     torch.cuda.reset_peak_memory_stats()
-    bench_unet()
+    bench_unet(channels_last=False)
+    torch.cuda.reset_peak_memory_stats()
+    bench_unet(channels_last=True)
     with open("gpurun_out/workloads.md", "w") as f:
         f.write("# Heavy-workload single-GPU measurements (MI355X)\n\n")
         for line in OUT:
