@@ -105,6 +105,8 @@ class FlServer:
         self.reports_manager.report(
             {"fit_start": str(start), "host_type": "server", "num_rounds": num_rounds}
         )
+        if hasattr(self.strategy, "num_rounds"):
+            self.strategy.num_rounds = num_rounds
         self.update_before_fit(num_rounds, timeout)
         self._get_initial_parameters(timeout)
 

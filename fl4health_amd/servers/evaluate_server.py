@@ -55,10 +55,11 @@ class EvaluateServer:
         start = datetime.datetime.now()
         result = self.federated_evaluate(timeout)
         end = datetime.datetime.now()
+        elapsed = (end - start).total_seconds()
         self.reports_manager.report(
-            {"fit_start": str(start), "fit_end": str(end), "fit_time_elapsed": round((end - start).total_seconds())}
+            {"fit_start": str(start), "fit_end": str(end), "fit_time_elapsed": round(elapsed)}
         )
-        return result
+        return result, elapsed
 
     def federated_evaluate(self, timeout: float | None = None) -> tuple[float | None, Metrics]:
         assert self.transport is not None, "EvaluateServer needs a transport (launch via simulation)"

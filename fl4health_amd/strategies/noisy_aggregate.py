@@ -30,7 +30,7 @@ def gaussian_noisy_unweighted_aggregate(
     n_slots = len(results[0][0].tensors)
     for slot in range(n_slots):
         summed = torch.stack([p.tensors[slot] for p, _ in results]).sum(dim=0)
-        out_tensors.append(_noisy_mean(summed, sigma, n_clients, seed, offset=slot * (1 << 20)))
+        out_tensors.append(_noisy_mean(summed, sigma, n_clients, seed, offset=slot * (1 << 40)))
     return Parameters(out_tensors, dict(results[0][0].meta))
 
 
@@ -54,7 +54,7 @@ def gaussian_noisy_weighted_aggregate(
     n_slots = len(results[0][0].tensors)
     for slot in range(n_slots):
         stack = torch.stack([p.tensors[slot] * c for (p, _), c in zip(results, coefs_scaled)])
-        out_tensors.append(_noisy_mean(stack.sum(dim=0), sigma, n_clients, seed, offset=slot * (1 << 20)))
+        out_tensors.append(_noisy_mean(stack.sum(dim=0), sigma, n_clients, seed, offset=slot * (1 << 40)))
     return Parameters(out_tensors, dict(results[0][0].meta))
 
 
