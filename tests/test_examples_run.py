@@ -1,0 +1,22 @@
+"""Examples stay runnable (1-round, tiny) — bitrot protection."""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+ROOT = Path(__file__).resolve().parent.parent
+
+EXAMPLES = ["basic_example", "fedprox_example", "scaffold_example", "fedpm_example"]
+
+
+@pytest.mark.parametrize("name", EXAMPLES)
+def test_example_runs(name):
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [sys.executable, "-m", f"examples.{name}.run", "--rounds", "1", "--local_steps", "1", "--batch_size", "8"],
+        capture_output=True, text=True, timeout=420, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "[SUMMARY]" in out.stdout
