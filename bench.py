@@ -45,6 +45,7 @@ class BenchFedProxClient(FedProxClient):
         if self.device.type == "cuda":
             self.autocast_dtype = torch.bfloat16
             self.use_cuda_graph = not args.no_graph
+            self.use_bf16_mirror = not args.no_mirror
 
     def get_model(self, config):
         model = ResNet18(num_classes=10)
@@ -118,6 +119,7 @@ def main() -> None:
     parser.add_argument("--batch_size", type=int, default=128)
     parser.add_argument("--shard_size", type=int, default=8192)
     parser.add_argument("--no_graph", action="store_true", help="disable hipGraph train-step capture")
+    parser.add_argument("--no_mirror", action="store_true", help="disable persistent bf16 weight mirrors")
     args = parser.parse_args()
 
     set_all_random_seeds(42)

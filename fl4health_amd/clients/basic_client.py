@@ -100,6 +100,10 @@ class BasicClient:
         self.use_cuda_graph = False
         self._graph: torch.cuda.CUDAGraph | None = None
         self._graph_static: dict[str, Any] | None = None
+        # persistent bf16 compute mirror for multi-dim params (autocast weight
+        # casts disappear; fused optimizer consumes bf16 grads + refreshes the
+        # mirror in one pass). Requires autocast training + flat optimizers.
+        self.use_bf16_mirror = False
 
         self.total_steps = 0
         self.total_epochs = 0
@@ -146,6 +150,8 @@ class BasicClient:
         from fl4health_amd.parameter_exchange.flat import FlatParameterView
 
         self.flat_view = FlatParameterView(self.model, bind=True)
+        if self.use_bf16_mirror and self.device.type == "cuda":
+            self.flat_view.enable_bf16_mirror()
         train_loader, val_loader = self.get_data_loaders(config)
         self.train_loader = train_loader
         self.val_loader = val_loader

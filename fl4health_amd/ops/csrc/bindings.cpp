@@ -5,8 +5,8 @@
 
 extern "C" {
 void launch_axpby(float*, const float*, float, float, int64_t, hipStream_t);
-void launch_prox_sgd(float*, const float*, const float*, float*, float, float, const float*,
-                     float, float, int, int64_t, hipStream_t);
+void launch_prox_sgd(float*, const void*, int, const float*, float*, unsigned short*, float,
+                     float, const float*, float, float, int, int64_t, hipStream_t);
 void launch_scaffold_sgd(float*, const float*, const float*, const float*, float, float, int64_t,
                          hipStream_t);
 void launch_scaffold_variate(float*, float*, const float*, const float*, const float*, float,
@@ -49,19 +49,28 @@ void axpby_(torch::Tensor y, torch::Tensor x, double a, double b) {
 }
 
 void prox_sgd_step_(torch::Tensor p, torch::Tensor g, c10::optional<torch::Tensor> w0,
-                    c10::optional<torch::Tensor> mbuf, double lr, double mu,
-                    c10::optional<torch::Tensor> mu_dev, double momentum, double weight_decay,
-                    bool nesterov) {
+                    c10::optional<torch::Tensor> mbuf, c10::optional<torch::Tensor> mirror,
+                    double lr, double mu, c10::optional<torch::Tensor> mu_dev, double momentum,
+                    double weight_decay, bool nesterov) {
   check_f32(p, "p");
-  check_f32(g, "g");
+  TORCH_CHECK(g.is_cuda() && g.is_contiguous(), "g must be contiguous GPU");
+  int grad_bf16 = g.scalar_type() == torch::kBFloat16 ? 1 : 0;
+  if (!grad_bf16) TORCH_CHECK(g.scalar_type() == torch::kFloat32, "g must be fp32 or bf16");
+  TORCH_CHECK(g.numel() == p.numel(), "grad size mismatch");
   const float* w0p = nullptr;
   float* mp = nullptr;
   const float* mudp = nullptr;
+  unsigned short* mirp = nullptr;
   if (w0.has_value()) { check_f32(*w0, "w0"); w0p = w0->data_ptr<float>(); }
   if (mbuf.has_value()) { check_f32(*mbuf, "mbuf"); mp = mbuf->data_ptr<float>(); }
   if (mu_dev.has_value()) { check_f32(*mu_dev, "mu_dev"); mudp = mu_dev->data_ptr<float>(); }
-  launch_prox_sgd(p.data_ptr<float>(), g.data_ptr<float>(), w0p, mp, (float)lr, (float)mu, mudp,
-                  (float)momentum, (float)weight_decay, nesterov ? 1 : 0, p.numel(), stream());
+  if (mirror.has_value()) {
+    TORCH_CHECK(mirror->scalar_type() == torch::kBFloat16 && mirror->numel() == p.numel());
+    mirp = reinterpret_cast<unsigned short*>(mirror->data_ptr());
+  }
+  launch_prox_sgd(p.data_ptr<float>(), g.data_ptr(), grad_bf16, w0p, mp, mirp, (float)lr,
+                  (float)mu, mudp, (float)momentum, (float)weight_decay, nesterov ? 1 : 0,
+                  p.numel(), stream());
 }
 
 void scaffold_sgd_step_(torch::Tensor p, torch::Tensor g, torch::Tensor c, torch::Tensor ci,

@@ -66,15 +66,32 @@ extern "C" void launch_axpby(float* y, const float* x, float a, float b, int64_t
 // (fl4health/losses/weight_drift_loss.py + torch SGD): one HBM pass.
 // w0 may be null (plain SGD). mbuf may be null (no momentum).
 // ---------------------------------------------------------------------------
+template <typename GT>
+__device__ __forceinline__ float load_grad(const GT* g, int64_t i);
+template <>
+__device__ __forceinline__ float load_grad<float>(const float* g, int64_t i) { return g[i]; }
+struct Bf16Tag { unsigned short v; };
+template <>
+__device__ __forceinline__ float load_grad<Bf16Tag>(const Bf16Tag* g, int64_t i) {
+  unsigned int bits = (unsigned int)g[i].v << 16;
+  return __uint_as_float(bits);
+}
+
+// fp32 master params; gradients may be fp32 OR bf16 (persistent bf16 weight
+// mirrors produce bf16 grads — the cast up happens inline here, fused).
+// Optional mirror output: after the update, p is re-cast bf16 into `mirror`
+// so the next forward reads fresh compute weights (one pass, no extra kernel).
+template <typename GT>
 __global__ __launch_bounds__(BLOCK) void prox_sgd_kernel(
-    float* __restrict__ p, const float* __restrict__ g, const float* __restrict__ w0,
-    float* __restrict__ mbuf, float lr, float mu, const float* __restrict__ mu_dev,
-    float momentum, float weight_decay, int nesterov, int64_t n) {
+    float* __restrict__ p, const GT* __restrict__ g, const float* __restrict__ w0,
+    float* __restrict__ mbuf, unsigned short* __restrict__ mirror, float lr, float mu,
+    const float* __restrict__ mu_dev, float momentum, float weight_decay, int nesterov,
+    int64_t n) {
   // mu may live in device memory (mu_dev) so a hipGraph-captured step sees
   // per-round server-adapted mu without re-capture
   if (mu_dev != nullptr) mu = mu_dev[0];
   GSL(i, n, STRIDE) {
-    float gi = g[i];
+    float gi = load_grad<GT>(g, i);
     float pi = p[i];
     if (weight_decay != 0.0f) gi = fmaf(weight_decay, pi, gi);
     if (w0 != nullptr) gi = fmaf(mu, pi - w0[i], gi);
@@ -84,15 +101,28 @@ __global__ __launch_bounds__(BLOCK) void prox_sgd_kernel(
       mbuf[i] = m;
       u = nesterov ? fmaf(momentum, m, gi) : m;
     }
-    p[i] = fmaf(-lr, u, pi);
+    float pnew = fmaf(-lr, u, pi);
+    p[i] = pnew;
+    if (mirror != nullptr) {
+      // round-to-nearest-even f32 -> bf16
+      unsigned int bits = __float_as_uint(pnew);
+      unsigned int rounded = bits + 0x7FFFu + ((bits >> 16) & 1u);
+      mirror[i] = (unsigned short)(rounded >> 16);
+    }
   }
 }
 
-extern "C" void launch_prox_sgd(float* p, const float* g, const float* w0, float* mbuf,
-                                float lr, float mu, const float* mu_dev, float momentum,
-                                float weight_decay, int nesterov, int64_t n, hipStream_t s) {
-  prox_sgd_kernel<<<grid_1d(n), BLOCK, 0, s>>>(p, g, w0, mbuf, lr, mu, mu_dev, momentum,
-                                               weight_decay, nesterov, n);
+extern "C" void launch_prox_sgd(float* p, const void* g, int grad_is_bf16, const float* w0,
+                                float* mbuf, unsigned short* mirror, float lr, float mu,
+                                const float* mu_dev, float momentum, float weight_decay,
+                                int nesterov, int64_t n, hipStream_t s) {
+  if (grad_is_bf16) {
+    prox_sgd_kernel<Bf16Tag><<<grid_1d(n), BLOCK, 0, s>>>(
+        p, (const Bf16Tag*)g, w0, mbuf, mirror, lr, mu, mu_dev, momentum, weight_decay, nesterov, n);
+  } else {
+    prox_sgd_kernel<float><<<grid_1d(n), BLOCK, 0, s>>>(
+        p, (const float*)g, w0, mbuf, mirror, lr, mu, mu_dev, momentum, weight_decay, nesterov, n);
+  }
 }
 
 // ---------------------------------------------------------------------------

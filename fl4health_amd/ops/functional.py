@@ -60,21 +60,23 @@ def prox_sgd_step_(
     momentum: float = 0.0,
     weight_decay: float = 0.0,
     nesterov: bool = False,
+    mirror: torch.Tensor | None = None,
 ) -> None:
     """Fused (proximal) SGD step: g' = g + wd*p + mu*(p - w0); momentum; p -= lr*u.
 
-    ``mu`` may be a device 0/1-dim fp32 tensor: the kernel reads it from device
-    memory, so a hipGraph-captured step tracks per-round mu adaptation.
+    ``mu`` may be a device 0/1-dim fp32 tensor (hipGraph-stable adaptation).
+    ``g`` may be bf16 (persistent bf16 weight mirrors); ``mirror`` (bf16, same
+    numel) receives the updated weights re-cast in the SAME kernel pass.
     """
     if p.is_cuda:
         _require_ext("prox_sgd_step_")
         if isinstance(mu, torch.Tensor):
-            _C.prox_sgd_step_(p, g, w0, mbuf, lr, 0.0, mu.reshape(1), momentum, weight_decay, nesterov)
+            _C.prox_sgd_step_(p, g, w0, mbuf, mirror, lr, 0.0, mu.reshape(1), momentum, weight_decay, nesterov)
         else:
-            _C.prox_sgd_step_(p, g, w0, mbuf, lr, mu, None, momentum, weight_decay, nesterov)
+            _C.prox_sgd_step_(p, g, w0, mbuf, mirror, lr, mu, None, momentum, weight_decay, nesterov)
         return
     mu_f = float(mu.item()) if isinstance(mu, torch.Tensor) else mu
-    geff = g.clone()
+    geff = g.float().clone()
     if weight_decay != 0.0:
         geff.add_(p, alpha=weight_decay)
     if w0 is not None and mu_f != 0.0:
@@ -84,6 +86,8 @@ def prox_sgd_step_(
         mbuf.mul_(momentum).add_(geff)
         u = geff.add(mbuf, alpha=momentum) if nesterov else mbuf
     p.add_(u, alpha=-lr)
+    if mirror is not None:
+        mirror.copy_(p.to(torch.bfloat16))
 
 
 def scaffold_sgd_step_(

@@ -28,8 +28,10 @@ class FlatOptimizerBase(Optimizer):
         super().__init__(params, defaults)
 
     def zero_grad(self, set_to_none: bool = False) -> None:  # noqa: ARG002
-        # grads are views of the flat buffer: zero in one pass, never detach
+        # grads are views of the flat buffer(s): zero in one pass, never detach
         self.gbuf.zero_()
+        if self.view.bf16_grad is not None:
+            self.view.bf16_grad.zero_()
 
     def state_dict(self) -> dict[str, Any]:
         sd = super().state_dict()
@@ -105,6 +107,29 @@ class FlatProxSGD(FlatOptimizerBase):
 
     @torch.no_grad()
     def step(self, closure=None) -> None:  # noqa: ARG002
+        if self.view.bf16_mirror is not None:
+            # region A: mirrored (multi-dim) params — bf16 grads consumed and
+            # updated weights re-cast into the mirror inside ONE kernel pass
+            m = self.view.mirror_numel
+            F.prox_sgd_step_(
+                self.view.params_region[:m],
+                self.view.bf16_grad,
+                self.w0[:m] if self.w0 is not None else None,
+                self.mbuf[:m] if self.mbuf is not None else None,
+                lr=self.lr, mu=self.mu, momentum=self.momentum,
+                weight_decay=self.weight_decay, nesterov=self.nesterov,
+                mirror=self.view.bf16_mirror,
+            )
+            # region B: 1D affine params (fp32 grads)
+            F.prox_sgd_step_(
+                self.view.params_region[m:],
+                self.gbuf[m:],
+                self.w0[m:] if self.w0 is not None else None,
+                self.mbuf[m:] if self.mbuf is not None else None,
+                lr=self.lr, mu=self.mu, momentum=self.momentum,
+                weight_decay=self.weight_decay, nesterov=self.nesterov,
+            )
+            return
         F.prox_sgd_step_(
             self.view.params_region,
             self.gbuf,

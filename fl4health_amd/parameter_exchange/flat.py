@@ -72,12 +72,22 @@ class FlatParameterView:
             # optimizer/penalty kernels (prox-SGD, SCAFFOLD correction, DP
             # clip) run over flat[:params_numel] in a single pass while
             # buffers (BN running stats...) are exchanged but never stepped.
-            param_names = [n for n, _ in module.named_parameters()]
+            # Within the params region, multi-dim (conv/linear) params come
+            # FIRST so an optional bf16 compute mirror covers one contiguous
+            # leading slice (1D affine params stay fp32 for the BN kernels).
+            all_params = [(n, p) for n, p in module.named_parameters()]
+            nd_names = [n for n, p in all_params if p.dim() >= 2]
+            oned_names = [n for n, p in all_params if p.dim() < 2]
+            param_names = nd_names + oned_names
             param_set = set(param_names)
             names = param_names + [n for n in sd.keys() if n not in param_set]
             self.params_numel = sum(sd[n].numel() for n in param_names)
+            self.mirror_numel = sum(sd[n].numel() for n in nd_names)
         else:
             self.params_numel = None  # unknown for custom subsets
+            self.mirror_numel = None
+        self.bf16_mirror: torch.Tensor | None = None
+        self.bf16_grad: torch.Tensor | None = None
         named = [(n, sd[n]) for n in names]
         self.spec = ParameterSpec.from_named_tensors(named)
         dev = device if device is not None else (named[0][1].device if named else "cpu")
@@ -94,6 +104,8 @@ class FlatParameterView:
         sd = self.module.state_dict()
         for i, name in enumerate(self.spec.names):
             t = sd[name]
+            if self.bf16_mirror is not None and t.dtype == torch.bfloat16:
+                continue  # fp32 master is authoritative for mirrored params
             dst = self.spec.slice_of(self.flat, i)
             if self.bound and dst.data_ptr() == t.data_ptr():
                 continue
@@ -105,6 +117,8 @@ class FlatParameterView:
         with torch.no_grad():
             for i, name in enumerate(self.spec.names):
                 t = sd[name]
+                if self.bf16_mirror is not None and t.dtype == torch.bfloat16:
+                    continue  # refreshed in one pass by sync_mirror_
                 src = self.spec.slice_of(self.flat, i)
                 if self.bound and src.data_ptr() == t.data_ptr():
                     continue
@@ -116,6 +130,7 @@ class FlatParameterView:
     def load_flat(self, flat: torch.Tensor) -> None:
         self.flat.copy_(flat.to(self.flat.device, torch.float32))
         self.push_into_module()
+        self.sync_mirror_()
 
     # ---- view binding (zero-copy hot path) -----------------------------
     def _bind_views(self) -> None:
@@ -159,6 +174,47 @@ class FlatParameterView:
         """Contiguous trainable-parameter slice of the flat buffer."""
         assert self.params_numel is not None, "params-first ordering required"
         return self.flat[: self.params_numel]
+
+    # ---- persistent bf16 compute mirror --------------------------------
+    def _strided_view(self, base: torch.Tensor, i: int, like: torch.Tensor) -> torch.Tensor:
+        n = int(torch.Size(self.spec.shapes[i]).numel())
+        sl = base[self.spec.offsets[i] : self.spec.offsets[i] + n]
+        p = like
+        if p.dim() == 4 and not p.is_contiguous() and p.is_contiguous(memory_format=torch.channels_last):
+            n_, c_, h_, w_ = p.shape
+            return sl.view(n_, h_, w_, c_).permute(0, 3, 1, 2)
+        return sl.view(p.shape)
+
+    def enable_bf16_mirror(self) -> None:
+        """Rebind multi-dim params as bf16 views of a persistent mirror buffer.
+
+        The fp32 master stays the source of truth (exchange, optimizer,
+        penalties); forwards read the bf16 mirror directly so autocast's
+        per-step weight-cast kernels disappear. The fused prox-SGD kernel
+        consumes the resulting bf16 grads and re-casts updated weights into
+        the mirror in the same pass. Requires bound views + autocast training.
+        """
+        assert self.bound and self.mirror_numel, "bind=True and multi-dim params required"
+        self.bf16_mirror = torch.empty(self.mirror_numel, dtype=torch.bfloat16, device=self.flat.device)
+        self.bf16_grad = torch.zeros(self.mirror_numel, dtype=torch.bfloat16, device=self.flat.device)
+        with torch.no_grad():
+            self.bf16_mirror.copy_(self.flat[: self.mirror_numel])
+        name_to_idx = {n: i for i, n in enumerate(self.spec.names)}
+        for mod_name, mod in self.module.named_modules():
+            prefix = mod_name + "." if mod_name else ""
+            for pname, p in list(mod.named_parameters(recurse=False)):
+                i = name_to_idx.get(prefix + pname)
+                if i is None or p.dim() < 2 or p.dtype != torch.float32:
+                    continue
+                view = self._strided_view(self.bf16_mirror, i, p)
+                new_p = nn.Parameter(view, requires_grad=p.requires_grad)
+                new_p.grad = self._strided_view(self.bf16_grad, i, p)
+                setattr(mod, pname, new_p)
+
+    def sync_mirror_(self) -> None:
+        if self.bf16_mirror is not None:
+            with torch.no_grad():
+                self.bf16_mirror.copy_(self.flat[: self.mirror_numel])
 
     def make_grad_buffer(self) -> torch.Tensor:
         """Allocate a flat grad buffer over the params region and point each

@@ -11,9 +11,11 @@ def test_params_first_layout_and_roundtrip():
     view = FlatParameterView(m)
     sd = m.state_dict()
     assert view.params_numel == sum(p.numel() for p in m.parameters())
-    # first names are exactly the parameter names in order
-    pnames = [n for n, _ in m.named_parameters()]
-    assert view.spec.names[: len(pnames)] == pnames
+    # params region = multi-dim params first, then 1D params (bf16-mirror layout)
+    nd = [n for n, p in m.named_parameters() if p.dim() >= 2]
+    oned = [n for n, p in m.named_parameters() if p.dim() < 2]
+    assert view.spec.names[: len(nd) + len(oned)] == nd + oned
+    assert view.mirror_numel == sum(p.numel() for _, p in m.named_parameters() if p.dim() >= 2)
     # roundtrip
     flat0 = view.clone_flat()
     for p in m.parameters():
@@ -83,3 +85,40 @@ def test_channels_last_binding_preserves_semantics():
         view.params_region.add_(gbuf, alpha=-0.1)
     for p, b, g in zip(m.parameters(), before, grads):
         assert torch.allclose(p.detach(), b - 0.1 * g, atol=1e-6)
+
+
+def test_bf16_mirror_semantics_cpu():
+    import torch
+
+    from fl4health_amd.ops import functional as F
+
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.BatchNorm2d(8), nn.Flatten(), nn.Linear(8 * 4 * 4, 5))
+    ref = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.BatchNorm2d(8), nn.Flatten(), nn.Linear(8 * 4 * 4, 5))
+    ref.load_state_dict(m.state_dict())
+
+    view = FlatParameterView(m, bind=True)
+    view.enable_bf16_mirror()
+    # multi-dim params are bf16 mirror views; 1D affine stay fp32 master views
+    assert m[0].weight.dtype == torch.bfloat16
+    assert m[1].weight.dtype == torch.float32
+    assert torch.allclose(m[0].weight.float(), ref[0].weight, atol=1e-2)
+
+    # one fused step with bf16 grads matches fp32 SGD at bf16 precision
+    g16 = torch.randn(view.mirror_numel).to(torch.bfloat16)
+    view.bf16_grad.copy_(g16)
+    master_before = view.params_region[: view.mirror_numel].clone()
+    F.prox_sgd_step_(
+        view.params_region[: view.mirror_numel], view.bf16_grad, None, None,
+        lr=0.1, mirror=view.bf16_mirror,
+    )
+    expected = master_before - 0.1 * g16.float()
+    assert torch.allclose(view.params_region[: view.mirror_numel], expected, atol=1e-6)
+    # mirror refreshed from updated master
+    assert torch.allclose(view.bf16_mirror.float(), expected, atol=1e-2)
+
+    # load_flat syncs the mirror
+    snap = view.clone_flat()
+    snap += 1.0
+    view.load_flat(snap)
+    assert torch.allclose(view.bf16_mirror.float(), snap[: view.mirror_numel], atol=3e-2)
