@@ -32,6 +32,7 @@ class Client(FedProxClient):
         if self.device.type == "cuda":
             self.autocast_dtype = torch.bfloat16
             self.use_cuda_graph = True
+            self.use_bf16_mirror = True
 
     def get_model(self, config):
         model = ResNet18(num_classes=10)
