@@ -475,9 +475,16 @@ class BasicClient:
     # ------------------------------------------------------------------
     def val_step(self, input: TorchInputType, target: TorchTargetType) -> tuple[EvaluationLosses, TorchPredType]:
         with torch.no_grad():
-            preds, features = self.predict(input)
-            target = self.transform_target(target)
-            losses = self.compute_evaluation_loss(preds, features, target)
+            if self.autocast_dtype is not None and self.device.type == "cuda":
+                # bf16-mirrored weights require autocast in eval as well
+                with torch.autocast(device_type="cuda", dtype=self.autocast_dtype):
+                    preds, features = self.predict(input)
+                    target = self.transform_target(target)
+                    losses = self.compute_evaluation_loss(preds, features, target)
+            else:
+                preds, features = self.predict(input)
+                target = self.transform_target(target)
+                losses = self.compute_evaluation_loss(preds, features, target)
         return losses, preds
 
     def _validate_on_loader(
