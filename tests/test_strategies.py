@@ -106,3 +106,41 @@ def test_collective_scales():
     sa = FedAvgWithAdaptiveConstraint(initial_parameters=init)
     scales = sa.collective_scales(1, 4, 2, 2)
     assert scales[0] == 0.25 and scales[1] == 0.5  # weighted model, unweighted loss
+
+
+def test_collective_aggregation_flags():
+    """Strategies whose aggregation is not a plain pre-scaled sum must force
+    the gather path in distributed mode."""
+    from fl4health_amd.common import Parameters
+    from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+    from fl4health_amd.strategies.client_dp_fedavgm import ClientLevelDPFedAvgM
+    from fl4health_amd.strategies.feddg_ga import FedDgGa
+    from fl4health_amd.strategies.fedavg_dynamic_layer import FedAvgDynamicLayer
+    from fl4health_amd.strategies.fedavg_sparse_coo_tensor import FedAvgSparseCooTensor
+    from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+    from fl4health_amd.strategies.fedopt import FedAdam
+    from fl4health_amd.strategies.fedpca import FedPCA
+    from fl4health_amd.strategies.fedpm import FedPm
+    from fl4health_amd.strategies.flash import Flash
+    from fl4health_amd.strategies.scaffold import Scaffold
+
+    init = Parameters([torch.zeros(4)])
+    collective = [
+        BasicFedAvg(),
+        FedAvgWithAdaptiveConstraint(initial_parameters=init),
+        Scaffold(initial_parameters=init),
+        FedAdam(initial_parameters=init),
+        Flash(initial_parameters=init),
+    ]
+    gather_only = [
+        ClientLevelDPFedAvgM(initial_parameters=init),
+        FedDgGa(),
+        FedPm(),
+        FedAvgDynamicLayer(),
+        FedAvgSparseCooTensor(),
+        FedPCA(),
+    ]
+    for s in collective:
+        assert s.supports_collective_aggregation(), type(s).__name__
+    for s in gather_only:
+        assert not s.supports_collective_aggregation(), type(s).__name__
