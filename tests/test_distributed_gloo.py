@@ -210,3 +210,70 @@ def test_distributed_partial_cohort(tmp_path):
     losses = json.loads(line[0][7:])["losses"]
     assert len(losses) == 2
     assert all(0 < l < 10 for _, l in losses)
+
+
+WORKER_FAULT = r"""
+import json, logging, torch
+logging.disable(logging.ERROR)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed
+from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+from tests.test_utils import TinyClient
+
+set_all_random_seeds(42)
+
+
+class FlakyClient(TinyClient):
+    """Fails fit on the FIRST round only (simulated transient client fault)."""
+
+    def fit(self, parameters, config):
+        if int(config["current_server_round"]) == 1 and self.client_name == "flaky":
+            raise RuntimeError("injected client fault")
+        return super().fit(parameters, config)
+
+
+def strategy_factory():
+    return BasicFedAvg(on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 2},
+                       min_fit_clients=1, min_evaluate_clients=1, min_available_clients=1)
+
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16},
+                    strategy_factory(), accept_failures=True)
+
+
+def client_factory(rank, world):
+    name = "flaky" if rank == 1 else f"ok{rank}"
+    return FlakyClient(seed=rank, metrics=[Accuracy()], device="cpu", client_name=name)
+
+
+hist = run_distributed(server_factory, client_factory, num_rounds=2,
+                       strategy_factory=strategy_factory, backend="gloo")
+if hist is not None:
+    print("RESULT " + json.dumps({"losses": hist.losses_distributed}))
+"""
+
+
+def test_distributed_client_fault_containment(tmp_path):
+    """A client exception on one rank must not deadlock the collectives; the
+    round aggregates over the surviving cohort (SURVEY 5.3 failure policy)."""
+    script = tmp_path / "worker_fault.py"
+    script.write_text(WORKER_FAULT)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29538",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    assert line, out.stdout[-2000:]
+    losses = json.loads(line[0][7:])["losses"]
+    assert len(losses) == 2  # both rounds completed despite the round-1 fault
