@@ -227,7 +227,7 @@ set_all_random_seeds(42)
 
 
 class FlakyClient(TinyClient):
-    """Fails fit on the FIRST round only (simulated transient client fault)."""
+    # Fails fit on the FIRST round only (simulated transient client fault).
 
     def fit(self, parameters, config):
         if int(config["current_server_round"]) == 1 and self.client_name == "flaky":
