@@ -29,14 +29,13 @@ log = logging.getLogger(__name__)
 
 
 class ClientLevelDPFedAvgM(BasicFedAvg):
-
-
     def supports_collective_aggregation(self) -> bool:
         # aggregation here is NOT a plain pre-scaled sum (noise/per-name/
         # posterior/SVD logic must see individual client payloads): force the
         # gather path so the distributed transport hands results to
         # aggregate_fit instead of all-reducing
         return False
+
     def __init__(
         self,
         *,

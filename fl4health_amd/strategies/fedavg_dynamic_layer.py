@@ -13,14 +13,13 @@ from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
 
 
 class FedAvgDynamicLayer(BasicFedAvg):
-
-
     def supports_collective_aggregation(self) -> bool:
         # aggregation here is NOT a plain pre-scaled sum (noise/per-name/
         # posterior/SVD logic must see individual client payloads): force the
         # gather path so the distributed transport hands results to
         # aggregate_fit instead of all-reducing
         return False
+
     def aggregate_fit(
         self,
         server_round: int,

@@ -12,14 +12,13 @@ from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
 
 
 class FedPCA(BasicFedAvg):
-
-
     def supports_collective_aggregation(self) -> bool:
         # aggregation here is NOT a plain pre-scaled sum (noise/per-name/
         # posterior/SVD logic must see individual client payloads): force the
         # gather path so the distributed transport hands results to
         # aggregate_fit instead of all-reducing
         return False
+
     def __init__(self, *, svd_merging: bool = True, **kwargs) -> None:
         kwargs.setdefault("weighted_aggregation", False)
         super().__init__(**kwargs)
