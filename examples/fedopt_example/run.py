@@ -1,0 +1,64 @@
+"""FedOpt example (capability of reference examples/fedopt_example): server-side
+adaptive optimizer (FedAdam / FedYogi / FedAdagrad / Flash) applied to the
+aggregated pseudo-gradient on the flat parameter buffer; the update itself runs
+in the fused server_opt HIP kernel on GPU."""
+from __future__ import annotations
+
+import torch
+
+from examples.common import example_argparser, initial_parameters, launch
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.clients.basic_client import BasicClient
+from fl4health_amd.datasets.synthetic import synthetic_cifar_loaders
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.models.cnn import SmallCnn
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.strategies.fedopt import FedAdagrad, FedAdam, FedYogi
+from fl4health_amd.strategies.flash import Flash
+
+STRATEGIES = {"adam": FedAdam, "yogi": FedYogi, "adagrad": FedAdagrad, "flash": Flash}
+
+
+class Client(BasicClient):
+    def __init__(self, seed: int, args, **kw) -> None:
+        super().__init__(**kw)
+        self.seed = seed
+        self.args = args
+
+    def get_model(self, config):
+        return SmallCnn()
+
+    def get_data_loaders(self, config):
+        return synthetic_cifar_loaders(n_train=1024, n_val=256, batch_size=self.args.batch_size, seed=self.seed)
+
+    def get_optimizer(self, config):
+        return torch.optim.SGD(self.model.parameters(), lr=0.05)
+
+    def get_criterion(self, config):
+        return torch.nn.CrossEntropyLoss()
+
+
+def main() -> None:
+    parser = example_argparser("FedOpt example")
+    parser.add_argument("--server_optimizer", choices=sorted(STRATEGIES), default="adam")
+    args = parser.parse_args()
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    strat_cls = STRATEGIES[args.server_optimizer]
+
+    def strategy_factory():
+        return strat_cls(
+            initial_parameters=initial_parameters(SmallCnn),
+            on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": args.local_steps},
+            min_fit_clients=1, min_evaluate_clients=1, min_available_clients=1,
+        )
+
+    def server_factory():
+        return FlServer(
+            SimpleClientManager(), {"n_server_rounds": args.rounds, "batch_size": args.batch_size}, strategy_factory()
+        )
+
+    launch(args, server_factory, lambda cid: Client(cid, args, metrics=[Accuracy()], device=device), strategy_factory)
+
+
+if __name__ == "__main__":
+    main()

@@ -8,7 +8,11 @@ import pytest
 
 ROOT = Path(__file__).resolve().parent.parent
 
-EXAMPLES = ["basic_example", "fedprox_example", "scaffold_example", "fedpm_example"]
+EXAMPLES = [
+    "basic_example", "fedprox_example", "scaffold_example", "fedpm_example",
+    "apfl_example", "moon_example", "fedopt_example", "feddg_ga_example",
+    "fedrep_example", "fenda_example", "model_merge_example", "dynamic_layer_exchange_example",
+]
 
 
 @pytest.mark.parametrize("name", EXAMPLES)
