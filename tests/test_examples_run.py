@@ -12,6 +12,7 @@ EXAMPLES = [
     "basic_example", "fedprox_example", "scaffold_example", "fedpm_example",
     "apfl_example", "moon_example", "fedopt_example", "feddg_ga_example",
     "fedrep_example", "fenda_example", "model_merge_example", "dynamic_layer_exchange_example",
+    "gpfl_example", "federated_eval_example", "ensemble_example", "mr_mtl_example", "perfcl_example",
 ]
 
 

@@ -539,3 +539,61 @@ def test_smoke_dp_scaffold():
     )
     hist = _run(server, clients)
     assert float(strategy.server_control_variates.abs().sum()) > 0
+
+
+def test_smoke_gpfl():
+    from fl4health_amd.clients.gpfl_client import GpflClient
+    from fl4health_amd.model_bases.gpfl_base import GpflModel
+    from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+
+    set_all_random_seeds(42)
+    feature_dim = 4 * 4 * 4
+
+    class Client(GpflClient, TinyClient):
+        def get_model(self, config):
+            base = nn.Sequential(
+                nn.Conv2d(3, 4, 3, padding=1), nn.ReLU(), nn.AdaptiveAvgPool2d(4), nn.Flatten()
+            )
+            return GpflModel(base, nn.Linear(feature_dim, 10), feature_dim, 10, flatten_features=False)
+
+        def get_optimizer(self, config):
+            return torch.optim.SGD(self.model.parameters(), lr=0.05)
+
+    clients = [Client(seed=i, n_train=N_TRAIN, metrics=[Accuracy()], device="cpu") for i in range(2)]
+    hist = _run(FlServer(SimpleClientManager(), CFG, BasicFedAvg(on_fit_config_fn=_fit_cfg)), clients)
+    assert hist is not None and len(hist.losses_distributed) == ROUNDS
+    # head stays personal: exchanged names exclude the prediction head
+    names = clients[0].parameter_exchanger.layers_to_transfer
+    assert all(not n.startswith("main_module.head_module") for n in names)
+
+
+def test_smoke_federated_evaluation(tmp_path):
+    """Evaluate-only FL: EvaluateServer + EvaluateClient, both a local
+    checkpoint model and the server-shipped global model are scored."""
+    from fl4health_amd.clients.evaluate_client import EvaluateClient
+    from fl4health_amd.servers.evaluate_server import EvaluateServer
+    from fl4health_amd.parallel.transports import InProcessClientProxy, InProcessTransport
+
+    set_all_random_seeds(42)
+    ckpt = tmp_path / "local_model.pt"
+    torch.save(TinyNet(), ckpt)
+    server_ckpt = tmp_path / "global_model.pt"
+    torch.save(TinyNet(), server_ckpt)
+
+    class Client(EvaluateClient, TinyClient):
+        pass
+
+    clients = [
+        Client(seed=i, n_train=N_TRAIN, metrics=[Accuracy()], device="cpu", model_checkpoint_path=ckpt)
+        for i in range(2)
+    ]
+    server = EvaluateServer(
+        SimpleClientManager(), model_checkpoint_path=server_ckpt, evaluate_config={"batch_size": 8}
+    )
+    transport = InProcessTransport(accept_failures=True)
+    server.transport = transport
+    for i, c in enumerate(clients):
+        server.client_manager.register(InProcessClientProxy(str(i), c))
+    (loss, metrics), _elapsed = server.fit()
+    assert loss is not None
+    assert any("local" in k for k in metrics) and any("global" in k for k in metrics)
