@@ -13,6 +13,7 @@ EXAMPLES = [
     "apfl_example", "moon_example", "fedopt_example", "feddg_ga_example",
     "fedrep_example", "fenda_example", "model_merge_example", "dynamic_layer_exchange_example",
     "gpfl_example", "federated_eval_example", "ensemble_example", "mr_mtl_example", "perfcl_example",
+    "fedsimclr_example", "ae_examples",
 ]
 
 
