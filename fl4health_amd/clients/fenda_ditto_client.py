@@ -56,7 +56,10 @@ class FendaDittoClient(BasicClient):
 
     def get_parameters(self, config: Config) -> Parameters:
         assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
-        model_params = self.parameter_exchanger.push_parameters(self.global_model, config=config)
+        # the exchanger's cached view covers the FENDA (personal) model; the
+        # exchanged payload is the Ditto GLOBAL model, so push via its own view
+        self.global_flat_view.pull_into_flat()
+        model_params = Parameters([self.global_flat_view.flat.detach().clone()])
         return self.parameter_exchanger.pack_parameters(model_params, self._vanilla_loss)
 
     def train_step(self, input, target):

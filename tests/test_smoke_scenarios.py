@@ -597,3 +597,54 @@ def test_smoke_federated_evaluation(tmp_path):
     (loss, metrics), _elapsed = server.fit()
     assert loss is not None
     assert any("local" in k for k in metrics) and any("global" in k for k in metrics)
+
+
+def test_smoke_fenda_ditto():
+    from fl4health_amd.clients.fenda_ditto_client import FendaDittoClient
+    from fl4health_amd.model_bases.fenda_base import FendaModel
+    from fl4health_amd.model_bases.parallel_split_models import ParallelFeatureJoinMode, ParallelSplitHeadModule
+    from fl4health_amd.model_bases.sequential_split_models import SequentiallySplitExchangeBaseModel
+    from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
+    from fl4health_amd.parameter_exchange.flat import FlatParameterView
+    from fl4health_amd.common import Parameters
+
+    class Head(ParallelSplitHeadModule):
+        def __init__(self):
+            super().__init__(ParallelFeatureJoinMode.CONCATENATE)
+            self.fc = nn.Linear(2 * 4 * 32 * 32, 10)
+
+        def parallel_output_join(self, local_tensor, global_tensor):
+            return torch.cat([local_tensor.flatten(1), global_tensor.flatten(1)], dim=1)
+
+        def head_forward(self, x):
+            return self.fc(x)
+
+    def extractor():
+        return nn.Sequential(nn.Conv2d(3, 4, 3, padding=1), nn.ReLU(), nn.Flatten())
+
+    def global_model():
+        return SequentiallySplitExchangeBaseModel(extractor(), nn.Linear(4 * 32 * 32, 10))
+
+    set_all_random_seeds(42)
+
+    class Client(FendaDittoClient, TinyClient):
+        def get_model(self, config):
+            return FendaModel(extractor(), extractor(), Head())
+
+        def get_global_model(self, config):
+            return global_model()
+
+        def get_optimizer(self, config):
+            return {"local": torch.optim.SGD(self.model.parameters(), lr=0.05), "global": None}
+
+        def setup_client(self, config):
+            super().setup_client(config)
+            self.optimizers["global"] = torch.optim.SGD(self.global_model.parameters(), lr=0.05)
+
+    clients = [Client(seed=i, n_train=N_TRAIN, metrics=[Accuracy()], device="cpu") for i in range(2)]
+    strategy = FedAvgWithAdaptiveConstraint(
+        initial_parameters=Parameters([FlatParameterView(global_model()).flat.clone()]),
+        initial_loss_weight=1.0, on_fit_config_fn=_fit_cfg,
+    )
+    hist = _run(FlServer(SimpleClientManager(), CFG, strategy), clients)
+    assert hist is not None and len(hist.losses_distributed) == ROUNDS
