@@ -14,6 +14,9 @@ EXAMPLES = [
     "fedrep_example", "fenda_example", "model_merge_example", "dynamic_layer_exchange_example",
     "gpfl_example", "federated_eval_example", "ensemble_example", "mr_mtl_example", "perfcl_example",
     "fedsimclr_example", "ae_examples",
+    "fenda_ditto_example", "feature_alignment_example", "fedpca_example", "warm_up_example",
+    "fl_plus_local_ft_example", "bert_finetuning_example", "sparse_tensor_partial_exchange_example",
+    "dp_scaffold_example",
 ]
 
 
