@@ -26,6 +26,10 @@ void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, 
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
 void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
                    float*, float, float, int64_t, int, int, int, int, hipStream_t);
+void launch_mkmmd_sums(const float*, const float*, double*, float*, int, int64_t, int64_t, int,
+                       hipStream_t);
+void launch_mkmmd_backward(const float*, const float*, const float*, float*, int, int64_t,
+                           int64_t, int, hipStream_t);
 void launch_bn_bwd(const void*, const void*, void*, float*, const float*, const float*,
                    const float*, const float*, float*, float*, float*, float*, int64_t, int, int,
                    int, int, hipStream_t);
@@ -262,6 +266,36 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tens
   return {dx, dgamma, dbeta};
 }
 
+// Fused multi-bandwidth Gaussian-kernel sums over a pairwise-distance Gram
+// (reference losses/mkmmd_loss.py:96-135).
+torch::Tensor mkmmd_sums(torch::Tensor d, torch::Tensor gammas, bool skip_diag) {
+  check_f32(d, "d");
+  check_f32(gammas, "gammas");
+  TORCH_CHECK(d.dim() == 2, "d must be [rows, cols]");
+  int k = (int)gammas.numel();
+  TORCH_CHECK(k >= 1 && k <= 32, "gammas must have 1..32 entries");
+  int64_t rows = d.size(0), cols = d.size(1);
+  if (skip_diag) TORCH_CHECK(rows == cols, "skip_diag needs a square Gram");
+  auto partial = torch::zeros({(int64_t)k * 512}, d.options().dtype(torch::kFloat64));
+  auto out = torch::empty({k}, d.options());
+  launch_mkmmd_sums(d.data_ptr<float>(), gammas.data_ptr<float>(), partial.data_ptr<double>(),
+                    out.data_ptr<float>(), k, rows, cols, skip_diag ? 1 : 0, stream());
+  return out;
+}
+
+torch::Tensor mkmmd_backward(torch::Tensor d, torch::Tensor gammas, torch::Tensor coef,
+                             bool skip_diag) {
+  check_f32(d, "d");
+  check_f32(gammas, "gammas");
+  check_f32(coef, "coef");
+  TORCH_CHECK(d.dim() == 2 && gammas.numel() == coef.numel(), "shape mismatch");
+  auto dd = torch::empty_like(d);
+  launch_mkmmd_backward(d.data_ptr<float>(), gammas.data_ptr<float>(), coef.data_ptr<float>(),
+                        dd.data_ptr<float>(), (int)gammas.numel(), d.size(0), d.size(1),
+                        skip_diag ? 1 : 0, stream());
+  return dd;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -280,4 +314,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("clip_rowsum_", &clip_rowsum_, "clipped per-sample grad sum");
   m.def("confusion_counts_", &confusion_counts_, "streaming TP/FP/FN/TN counts");
   m.def("weighted_sum_rows", &weighted_sum_rows, "out = sum_k w[k]*stack[k]");
+  m.def("mkmmd_sums", &mkmmd_sums, "per-bandwidth Gaussian kernel sums over a Gram");
+  m.def("mkmmd_backward", &mkmmd_backward, "dL/dGram for mkmmd_sums");
 }

@@ -292,3 +292,30 @@ def weighted_sum_rows(stack: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
 def noise_multiplier_sigma(noise_multiplier: float, clip_bound: float, num: int) -> float:
     """Std-dev of per-coordinate Gaussian noise for a summed, clipped aggregate."""
     return noise_multiplier * clip_bound / max(num, 1) * math.sqrt(1.0)
+
+
+def mkmmd_sums(d: torch.Tensor, gammas: torch.Tensor, skip_diag: bool) -> torch.Tensor:
+    """Per-bandwidth Gaussian kernel sums over a pairwise-distance Gram:
+    out[k] = sum_{ij (i != j if skip_diag)} exp(-gammas[k] * d[i, j]).
+    GPU: one fused pass (mmd_ops.hip); CPU: torch oracle."""
+    if d.is_cuda:
+        _require_ext("mkmmd_sums")
+        return _C.mkmmd_sums(d, gammas, skip_diag)
+    k = torch.exp(-d.unsqueeze(0) * gammas.reshape(-1, 1, 1))
+    if skip_diag:
+        eye = torch.eye(d.shape[0], dtype=torch.bool, device=d.device)
+        k = k.masked_fill(eye, 0.0)
+    return k.sum(dim=(1, 2))
+
+
+def mkmmd_sums_backward(d: torch.Tensor, gammas: torch.Tensor, coef: torch.Tensor,
+                        skip_diag: bool) -> torch.Tensor:
+    """dL/d(Gram) for mkmmd_sums given upstream per-bandwidth grads `coef`."""
+    if d.is_cuda:
+        _require_ext("mkmmd_backward")
+        return _C.mkmmd_backward(d, gammas, coef, skip_diag)
+    dd = (coef.reshape(-1, 1, 1) * (-gammas.reshape(-1, 1, 1)) * torch.exp(-d.unsqueeze(0) * gammas.reshape(-1, 1, 1))).sum(dim=0)
+    if skip_diag:
+        eye = torch.eye(d.shape[0], dtype=torch.bool, device=d.device)
+        dd = dd.masked_fill(eye, 0.0)
+    return dd

@@ -23,6 +23,7 @@ setup(
                 os.path.join(CSRC, "bindings.cpp"),
                 os.path.join(CSRC, "flat_ops.hip"),
                 os.path.join(CSRC, "bn_ops.hip"),
+                os.path.join(CSRC, "mmd_ops.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3"],
