@@ -277,3 +277,72 @@ def test_distributed_client_fault_containment(tmp_path):
     assert line, out.stdout[-2000:]
     losses = json.loads(line[0][7:])["losses"]
     assert len(losses) == 2  # both rounds completed despite the round-1 fault
+
+
+WORKER_GATHER = r"""
+import json, logging, torch
+logging.disable(logging.INFO)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.clients.partial_weight_exchange_client import PartialWeightExchangeClient
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed, run_simulation
+from fl4health_amd.strategies.fedavg_dynamic_layer import FedAvgDynamicLayer
+from tests.test_utils import TinyClient
+
+
+class Client(PartialWeightExchangeClient, TinyClient):
+    pass
+
+
+def strategy_factory():
+    return FedAvgDynamicLayer(on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 3})
+
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 2, "batch_size": 16}, strategy_factory())
+
+
+def client_factory(rank, world):
+    return Client(seed=rank, exchange_percentage=0.5, metrics=[Accuracy()], device="cpu")
+
+
+# distributed run: the strategy forces the GATHER path (meta-carrying
+# Parameters move over all_gather_object, not the pre-scaled all-reduce)
+set_all_random_seeds(42)
+hist = run_distributed(server_factory, client_factory, num_rounds=2,
+                       strategy_factory=strategy_factory, backend="gloo")
+if hist is not None:
+    # identical in-process simulation for cross-checking
+    set_all_random_seeds(42)
+    sim_clients = [client_factory(i, 2) for i in range(2)]
+    sim_hist = run_simulation(server_factory(), sim_clients, num_rounds=2)
+    print("RESULT " + json.dumps({
+        "dist": hist.losses_distributed, "sim": sim_hist.losses_distributed}))
+"""
+
+
+def test_distributed_gather_path_matches_simulation(tmp_path):
+    """FedAvgDynamicLayer (supports_collective_aggregation() == False) must go
+    through the object-gather path in distributed mode and produce the same
+    per-round losses as the in-process simulation."""
+    script = tmp_path / "worker_gather.py"
+    script.write_text(WORKER_GATHER)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29539",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    assert line, out.stdout[-2000:]
+    res = json.loads(line[0][7:])
+    assert len(res["dist"]) == 2
+    for (rd, ld), (rs, ls) in zip(res["dist"], res["sim"]):
+        assert rd == rs and abs(ld - ls) < 5e-4, (res["dist"], res["sim"])
