@@ -241,3 +241,87 @@ def test_device_tensor_loader_epochs():
     e1 = torch.cat([yy for _, yy in loader])
     e2 = torch.cat([yy for _, yy in loader])
     assert not torch.equal(e1, e2)
+
+
+def test_typed_datasets_and_ssl_pairs():
+    from fl4health_amd.utils.dataset import (
+        DictionaryDataset, SslTensorDataset, TensorDataset, select_by_indices,
+    )
+
+    x = torch.arange(24, dtype=torch.float32).reshape(6, 4)
+    y = torch.arange(6)
+    ds = TensorDataset(x, y, transform=lambda v: v * 2)
+    xi, yi = ds[1]
+    assert torch.equal(xi, x[1] * 2) and yi == 1
+    ds.update_transform(lambda v: v + 1)  # composes: (v*2)+1
+    xi, _ = ds[1]
+    assert torch.equal(xi, x[1] * 2 + 1)
+
+    sub = select_by_indices(ds, torch.tensor([0, 2]))
+    assert len(sub) == 2 and sub[1][1] == 2
+
+    ssl = SslTensorDataset(x, target_transform=lambda v: -v)
+    xi, view = ssl[3]
+    assert torch.equal(view, -x[3]) and torch.equal(xi, x[3])
+
+    dd = DictionaryDataset({"a": [x[i] for i in range(6)], "b": [y[i] for i in range(6)]}, y)
+    item, target = dd[2]
+    assert set(item) == {"a", "b"} and target == 2
+    assert len(dd) == 6
+
+
+def test_autoencoder_dataset_converter_roundtrip():
+    from fl4health_amd.model_bases.autoencoders_base import ConditionalVae
+    from fl4health_amd.utils.dataset import TensorDataset
+    from fl4health_amd.utils.dataset_converter import AutoEncoderDatasetConverter
+
+    x = torch.randn(10, 3, 4)
+    y = torch.randint(0, 5, (10,))
+    conv = AutoEncoderDatasetConverter(condition="label").convert_dataset(TensorDataset(x, y))
+    packed, target = conv[0]
+    assert packed.numel() == 12 + 5  # flat input + one-hot(5)
+    assert torch.equal(target, x[0])
+    unpack = conv.get_unpacking_function()
+    batch = torch.stack([conv[i][0] for i in range(4)])
+    xr, cond = unpack(batch)
+    assert xr.shape == (4, 3, 4) and cond.shape == (4, 5)
+    assert torch.allclose(xr[0], x[0])
+
+    # unconditioned: target == input, unpack is identity-shaped
+    conv2 = AutoEncoderDatasetConverter(condition=None).convert_dataset(TensorDataset(x, y))
+    p2, t2 = conv2[3]
+    assert torch.equal(p2, x[3]) and torch.equal(t2, x[3])
+
+    # drives a CVAE end to end
+    class Enc(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.mu = nn.Linear(12 + 5, 2)
+            self.logvar = nn.Linear(12 + 5, 2)
+
+        def forward(self, xx, cond):
+            h = torch.cat([xx.flatten(1), cond], dim=1)
+            return self.mu(h), self.logvar(h)
+
+    class Dec(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(2 + 5, 12)
+
+        def forward(self, z, cond):
+            return self.fc(torch.cat([z, cond], dim=1))
+
+    cvae = ConditionalVae(Enc(), Dec(), unpack_input_condition=unpack)
+    out = cvae(batch)
+    assert out.shape == (4, 12 + 2 + 2)
+
+
+def test_msd_registry_and_logging_mode():
+    from fl4health_amd.utils.logging import LoggingMode
+    from fl4health_amd.utils.msd_dataset_sources import MSD_TASK_DIMS, MsdDataset, get_msd_dataset_enum
+
+    assert get_msd_dataset_enum("Task09_Spleen") == MsdDataset.TASK09_SPLEEN
+    assert MSD_TASK_DIMS[MsdDataset.TASK01_BRAINTUMOUR] == (4, 4)
+    with pytest.raises(ValueError):
+        get_msd_dataset_enum("nope")
+    assert LoggingMode.VALIDATION.value == "Validation"
