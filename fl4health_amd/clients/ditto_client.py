@@ -47,9 +47,29 @@ class DittoClient(BasicClient):
         self.global_model = self.get_global_model(config)
         self.global_flat_view = FlatParameterView(self.global_model, bind=True)
         self.global_model.train()
+        if set(self.optimizers) == {"global"}:
+            # single-optimizer form (reference mixins/personalized/ditto.py:95):
+            # the returned optimizer drives the PERSONAL model; clone its
+            # configuration for the global model
+            personal = self.optimizers["global"]
+            self.optimizers = {"local": personal, "global": self._clone_optimizer_for_global(personal)}
         assert "global" in self.optimizers and "local" in self.optimizers, (
             "DittoClient requires get_optimizer to return {'global': ..., 'local': ...}"
         )
+
+    def _clone_optimizer_for_global(self, personal: torch.optim.Optimizer) -> torch.optim.Optimizer:
+        from fl4health_amd.optimizers import FlatProxSGD
+
+        group = personal.param_groups[0]
+        if isinstance(personal, FlatProxSGD):
+            return FlatProxSGD(
+                self.global_flat_view,
+                lr=group["lr"],
+                momentum=group.get("momentum", 0.0),
+                weight_decay=group.get("weight_decay", 0.0),
+            )
+        defaults = dict(personal.defaults)
+        return type(personal)(self.global_model.parameters(), **defaults)
 
     # ------------------------------------------------------------------
     def set_parameters(self, parameters: Parameters, config: Config, fitting_round: bool) -> None:
@@ -72,7 +92,10 @@ class DittoClient(BasicClient):
             # round-0 initialization handshake: full weights, no aux
             return self.setup_client_and_return_all_model_parameters(config)
         assert isinstance(self.parameter_exchanger, FullParameterExchangerWithPacking)
-        model_params = self.parameter_exchanger.push_parameters(self.global_model, config=config)
+        # the exchanger's cached view covers the PERSONAL model; Ditto
+        # exchanges the GLOBAL model, so push through its own flat view
+        self.global_flat_view.pull_into_flat()
+        model_params = Parameters([self.global_flat_view.flat.detach().clone()])
         return self.parameter_exchanger.pack_parameters(model_params, self._vanilla_loss_for_packing)
 
     # ------------------------------------------------------------------
