@@ -39,11 +39,16 @@ class DpSgdEngine:
     @torch.no_grad()
     def step(self) -> None:
         params = [p for p in self.module.per_sample_params() if getattr(p, "grad_sample", None) is not None]
-        if not params:
+        ghost = list(getattr(self.module, "_ghost", {}).items())
+        if not params and not ghost:
             self.optimizer.step()
             return
-        device = params[0].device
-        batch = params[0].grad_sample.shape[0]
+        if params:
+            device = params[0].device
+            batch = params[0].grad_sample.shape[0]
+        else:
+            device = ghost[0][1][0].device
+            batch = ghost[0][1][0].shape[0]
         # With mean-reduced losses the captured grad_sample[b] is (1/B) dL_b,
         # i.e. already carries the 1/B factor: clip at C/B (the coefficient
         # min(1, (C/B)/(||dL_b||/B)) equals the true min(1, C/||dL_b||)), skip
@@ -59,22 +64,42 @@ class DpSgdEngine:
         sqnorms = torch.zeros(batch, dtype=torch.float32, device=device)
         for p in params:
             F.per_sample_sqnorm_(p.grad_sample.reshape(batch, -1).float(), sqnorms)
+        for _m, (a, g) in ghost:
+            gf, af = g.float(), a.float()
+            g_sq = gf.pow(2).sum(dim=1)
+            # ||g_b (x) a_b||_F^2 = ||g_b||^2 * ||a_b||^2 (weight) + ||g_b||^2 (bias)
+            sqnorms += g_sq * af.pow(2).sum(dim=1)
+            if _m.bias is not None:
+                sqnorms += g_sq
+        coef = torch.clamp(eff_bound / (sqnorms.sqrt() + 1e-6), max=1.0)
+        for _m, (a, g) in ghost:
+            cg = coef.unsqueeze(1) * g.float()  # [B, out]
+            gw = cg.t() @ a.float()  # clipped per-sample sum as ONE GEMM
+            self._finalize_grad(_m.weight, gw.reshape(-1), sigma, final_div)
+            if _m.bias is not None:
+                self._finalize_grad(_m.bias, cg.sum(dim=0), sigma, final_div)
+        if ghost:
+            self.module._ghost.clear()
         for p in params:
             g = torch.zeros(p.numel(), dtype=torch.float32, device=device)
             F.clip_rowsum_(p.grad_sample.reshape(batch, -1).float(), sqnorms, g, eff_bound)
-            if self.noise_multiplier > 0:
-                F.gaussian_noise_(g, sigma=sigma, seed=self.seed, offset=self._noise_counter)
-                self._noise_counter += (p.numel() + 3) // 4 + 1
-            if final_div != 1.0:
-                g /= final_div
-            # write INTO existing .grad when present (it may alias a flat grad
-            # buffer used by fused optimizers) rather than rebinding
-            if p.grad is not None and p.grad.shape == p.shape:
-                p.grad.copy_(g.view(p.shape).to(p.grad.dtype))
-            else:
-                p.grad = g.view(p.shape).to(p.dtype)
+            self._finalize_grad(p, g, sigma, final_div)
             p.grad_sample = None
         self.optimizer.step()
+
+    def _finalize_grad(self, p: torch.Tensor, g: torch.Tensor, sigma: float, final_div: float) -> None:
+        """Noise + scale the clipped sum and write it into p.grad (in place
+        when .grad aliases a flat buffer used by fused optimizers)."""
+        g = g.reshape(-1).contiguous()
+        if self.noise_multiplier > 0:
+            F.gaussian_noise_(g, sigma=sigma, seed=self.seed, offset=self._noise_counter)
+            self._noise_counter += (p.numel() + 3) // 4 + 1
+        if final_div != 1.0:
+            g /= final_div
+        if p.grad is not None and p.grad.shape == p.shape:
+            p.grad.copy_(g.view(p.shape).to(p.grad.dtype))
+        else:
+            p.grad = g.view(p.shape).to(p.dtype)
 
     def zero_grad(self, set_to_none: bool = False) -> None:
         self.module.clear_grad_samples()

@@ -41,13 +41,20 @@ class GradSampleModule(nn.Module):
     """Wraps a module; after backward, each supported layer's parameters carry
     .grad_sample of shape [B, *param_shape]."""
 
-    def __init__(self, module: nn.Module) -> None:
+    def __init__(self, module: nn.Module, ghost_clipping: bool = True) -> None:
         super().__init__()
         validate_module(module)
         self._module = module
         self._hooks: list = []
         self._activations: dict[nn.Module, torch.Tensor] = {}
         self.hooks_enabled = True
+        # Ghost clipping (Linear, single-token): ||g_b (x) a_b||_F = ||g_b||*||a_b||,
+        # so the per-sample norm needs only the captured (act, grad_out) pair and
+        # the clipped sum is ONE rocBLAS GEMM (c_b*g_b)^T A — the [B, out, in]
+        # per-sample grad tensor is never materialized (for a CNN's big FC layer
+        # that is ~B*|W|*4 bytes of HBM traffic per step saved).
+        self.ghost_clipping = ghost_clipping
+        self._ghost: dict[nn.Module, tuple[torch.Tensor, torch.Tensor]] = {}
         self._register_hooks()
 
     @property
@@ -91,6 +98,16 @@ class GradSampleModule(nn.Module):
 
     def _compute_grad_sample(self, m: nn.Module, act: torch.Tensor, go: torch.Tensor) -> None:
         if isinstance(m, nn.Linear):
+            if self.ghost_clipping and act.dim() == 2 and m not in self._ghost and m.weight.requires_grad:
+                self._ghost[m] = (act, go)
+                return
+            if m in self._ghost:
+                # module fired more than once this step: materialize the stored
+                # pair and fall through to the per-sample-grad path for both
+                a0, g0 = self._ghost.pop(m)
+                self._store(m.weight, torch.einsum("bo,bi->boi", g0, a0))
+                if m.bias is not None:
+                    self._store(m.bias, g0)
             a2 = act.reshape(act.shape[0], -1, act.shape[-1])  # [B, T, in]
             g2 = go.reshape(go.shape[0], -1, go.shape[-1])  # [B, T, out]
             self._store(m.weight, torch.einsum("bto,bti->boi", g2, a2))
@@ -146,6 +163,7 @@ class GradSampleModule(nn.Module):
         for p in self._module.parameters():
             if hasattr(p, "grad_sample"):
                 p.grad_sample = None
+        self._ghost.clear()
 
     def per_sample_params(self) -> list[torch.Tensor]:
         return [p for p in self._module.parameters() if p.requires_grad]

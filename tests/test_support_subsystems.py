@@ -325,3 +325,61 @@ def test_msd_registry_and_logging_mode():
     with pytest.raises(ValueError):
         get_msd_dataset_enum("nope")
     assert LoggingMode.VALIDATION.value == "Validation"
+
+
+def test_ghost_clipping_matches_materialized():
+    """Ghost-clipped DP-SGD (per-sample norms via ||g||*||a||, clipped sum as
+    one GEMM) must match the materialized per-sample-grad path exactly when
+    noise is off (noise streams differ only in parameter ordering)."""
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    def run(ghost):
+        set_all_random_seeds(0)
+        model = nn.Sequential(nn.Flatten(), nn.Linear(12, 16), nn.ReLU(), nn.Linear(16, 4))
+        gsm = GradSampleModule(model, ghost_clipping=ghost)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        eng = DpSgdEngine(gsm, opt, noise_multiplier=0.0, clipping_bound=0.9, seed=5)
+        x = torch.randn(8, 3, 2, 2)
+        y = torch.randint(0, 4, (8,))
+        for _ in range(3):
+            eng.zero_grad()
+            nn.functional.cross_entropy(gsm(x), y).backward()
+            eng.step()
+        return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+
+    assert torch.allclose(run(True), run(False), atol=1e-6)
+
+
+def test_ghost_clipping_mixed_with_conv_and_repeat_fire():
+    """Conv layers stay on the materialized path; a Linear that fires twice in
+    one step must fall back to materialization (weight-shared/recurrent use)."""
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    class TwiceNet(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.conv = nn.Conv2d(1, 2, 3, padding=1)
+            self.fc = nn.Linear(8, 8)
+            self.head = nn.Linear(8, 3)
+
+        def forward(self, x):
+            h = self.conv(x).flatten(1)
+            h = self.fc(torch.relu(self.fc(h)))  # fc fires TWICE
+            return self.head(h)
+
+    def run(ghost):
+        set_all_random_seeds(1)
+        model = TwiceNet()
+        gsm = GradSampleModule(model, ghost_clipping=ghost)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        eng = DpSgdEngine(gsm, opt, noise_multiplier=0.0, clipping_bound=0.5, seed=2)
+        x = torch.randn(6, 1, 2, 2)
+        y = torch.randint(0, 3, (6,))
+        eng.zero_grad()
+        nn.functional.cross_entropy(gsm(x), y).backward()
+        eng.step()
+        return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+
+    assert torch.allclose(run(True), run(False), atol=1e-6)
