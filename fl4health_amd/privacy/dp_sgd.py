@@ -64,20 +64,41 @@ class DpSgdEngine:
         sqnorms = torch.zeros(batch, dtype=torch.float32, device=device)
         for p in params:
             F.per_sample_sqnorm_(p.grad_sample.reshape(batch, -1).float(), sqnorms)
+        prepared = []
         for _m, (a, g) in ghost:
-            gf, af = g.float(), a.float()
-            g_sq = gf.pow(2).sum(dim=1)
-            # ||g_b (x) a_b||_F^2 = ||g_b||^2 * ||a_b||^2 (weight) + ||g_b||^2 (bias)
-            sqnorms += g_sq * af.pow(2).sum(dim=1)
-            if _m.bias is not None:
-                sqnorms += g_sq
+            if isinstance(_m, torch.nn.Linear):
+                gf, af = g.float(), a.float()
+                g_sq = gf.pow(2).sum(dim=1)
+                # ||g_b (x) a_b||_F^2 = ||g_b||^2 * ||a_b||^2 (weight) + ||g_b||^2 (bias)
+                sqnorms += g_sq * af.pow(2).sum(dim=1)
+                if _m.bias is not None:
+                    sqnorms += g_sq
+                prepared.append((_m, "linear", af, gf))
+            else:  # Conv2d ghost-norm: ||A U^T||_F^2 = <U^T U, A^T A> per sample
+                u = torch.nn.functional.unfold(
+                    a, _m.kernel_size, _m.dilation, _m.padding, _m.stride
+                ).float()  # [B, D, L]
+                g2 = g.reshape(g.shape[0], g.shape[1], -1).float()  # [B, O, L]
+                gram_u = torch.bmm(u.transpose(1, 2), u)  # [B, L, L]
+                gram_g = torch.bmm(g2.transpose(1, 2), g2)  # [B, L, L]
+                sqnorms += (gram_u * gram_g).sum(dim=(1, 2))
+                if _m.bias is not None:
+                    sqnorms += g2.sum(dim=2).pow(2).sum(dim=1)
+                prepared.append((_m, "conv", u, g2))
         coef = torch.clamp(eff_bound / (sqnorms.sqrt() + 1e-6), max=1.0)
-        for _m, (a, g) in ghost:
-            cg = coef.unsqueeze(1) * g.float()  # [B, out]
-            gw = cg.t() @ a.float()  # clipped per-sample sum as ONE GEMM
-            self._finalize_grad(_m.weight, gw.reshape(-1), sigma, final_div)
-            if _m.bias is not None:
-                self._finalize_grad(_m.bias, cg.sum(dim=0), sigma, final_div)
+        for _m, kind, a_or_u, gf in prepared:
+            if kind == "linear":
+                cg = coef.unsqueeze(1) * gf  # [B, out]
+                gw = cg.t() @ a_or_u  # clipped per-sample sum as ONE GEMM
+                self._finalize_grad(_m.weight, gw.reshape(-1), sigma, final_div)
+                if _m.bias is not None:
+                    self._finalize_grad(_m.bias, cg.sum(dim=0), sigma, final_div)
+            else:
+                cg = coef.view(-1, 1, 1) * gf  # [B, O, L]
+                gw = torch.einsum("bol,bdl->od", cg, a_or_u)  # one contraction over (b, l)
+                self._finalize_grad(_m.weight, gw.reshape(-1), sigma, final_div)
+                if _m.bias is not None:
+                    self._finalize_grad(_m.bias, cg.sum(dim=(0, 2)), sigma, final_div)
         if ghost:
             self.module._ghost.clear()
         for p in params:

@@ -28,11 +28,15 @@ def validate_module(module: nn.Module) -> None:
 
 
 def convert_batchnorm_modules(model: nn.Module) -> nn.Module:
-    """Replace BatchNorm with GroupNorm (reference privacy_utilities.py:44-71)."""
+    """Replace BatchNorm with GroupNorm (reference privacy_utilities.py:44-71).
+    Also disables in-place activations: full backward hooks (the per-sample
+    grad capture) forbid in-place mutation of their outputs."""
     for name, child in list(model.named_children()):
         if isinstance(child, (nn.BatchNorm1d, nn.BatchNorm2d, nn.BatchNorm3d)):
             setattr(model, name, nn.GroupNorm(min(32, child.num_features), child.num_features, affine=True))
         else:
+            if getattr(child, "inplace", False):
+                child.inplace = False
             convert_batchnorm_modules(child)
     return model
 
@@ -115,6 +119,19 @@ class GradSampleModule(nn.Module):
                 self._store(m.bias, g2.sum(dim=1))
         elif isinstance(m, (nn.Conv1d, nn.Conv2d)):
             b = act.shape[0]
+            if (
+                self.ghost_clipping
+                and isinstance(m, nn.Conv2d)
+                and m.groups == 1
+                and m not in self._ghost
+            ):
+                # ghost-norm pays when the [L, L] Grams are smaller than the
+                # per-sample grad itself: L^2 < |W| (deep ResNet blocks: L=64,
+                # |W|=590k). Large-spatial early convs stay materialized.
+                l_spatial = go.shape[2:].numel()
+                if l_spatial * l_spatial < m.weight.numel():
+                    self._ghost[m] = (act, go)
+                    return
             if isinstance(m, nn.Conv2d):
                 unfolded = Fn.unfold(act, m.kernel_size, m.dilation, m.padding, m.stride)  # [B, Cin*k*k, L]
             else:

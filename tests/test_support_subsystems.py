@@ -383,3 +383,26 @@ def test_ghost_clipping_mixed_with_conv_and_repeat_fire():
         return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
 
     assert torch.allclose(run(True), run(False), atol=1e-6)
+
+
+def test_ghost_clipping_conv2d_matches_materialized():
+    """Conv2d ghost-norm (<U^T U, A^T A> Gram trick) engages when L^2 < |W|
+    and must match the materialized einsum path exactly (zero noise)."""
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    def run(ghost):
+        set_all_random_seeds(0)
+        model = nn.Sequential(nn.Conv2d(64, 64, 3, padding=1), nn.ReLU(), nn.Flatten(), nn.Linear(64 * 4 * 4, 5))
+        gsm = GradSampleModule(model, ghost_clipping=ghost)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        eng = DpSgdEngine(gsm, opt, noise_multiplier=0.0, clipping_bound=0.7, seed=3)
+        x = torch.randn(6, 64, 4, 4)
+        y = torch.randint(0, 5, (6,))
+        for _ in range(2):
+            eng.zero_grad()
+            nn.functional.cross_entropy(gsm(x), y).backward()
+            eng.step()
+        return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+
+    assert torch.allclose(run(True), run(False), atol=1e-5)

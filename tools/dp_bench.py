@@ -48,6 +48,7 @@ def main():
     # materializing the [B, out, in] per-sample grads of the Linear layers
     results = {}
     for ghost in (False, True):
+        print(f"smallcnn ghost={ghost}...", flush=True)
         model2 = convert_batchnorm_modules(SmallCnn()).cuda()
         gsm = GradSampleModule(model2, ghost_clipping=ghost)
         opt2 = torch.optim.SGD(model2.parameters(), lr=0.05)
@@ -66,6 +67,31 @@ def main():
         f"DP-SGD materialized {results[False]:.3f} ms ({results[False] / t_plain:.2f}x) | "
         f"DP-SGD ghost {results[True]:.3f} ms ({results[True] / t_plain:.2f}x, "
         f"{batch * 1000 / results[True]:.0f} private samples/s)"
+    )
+
+    # flagship-model DP: ResNet-18 (GN), where deep blocks hit the conv
+    # ghost-norm path (L^2 << |W|)
+    from fl4health_amd.models.resnet import ResNet18
+
+    res = {}
+    for ghost in (False, True):
+        print(f"resnet ghost={ghost}...", flush=True)
+        model3 = convert_batchnorm_modules(ResNet18(num_classes=10)).cuda()
+        gsm = GradSampleModule(model3, ghost_clipping=ghost)
+        opt3 = torch.optim.SGD(model3.parameters(), lr=0.05)
+        engine = DpSgdEngine(gsm, opt3, noise_multiplier=1.0, clipping_bound=1.0, seed=0)
+
+        def dp3():
+            engine.zero_grad()
+            criterion(gsm(x), y).backward()
+            engine.step()
+
+        res[ghost] = bench_step(dp3, iters=30, warmup=5)
+    n3 = sum(p.numel() for p in model3.parameters())
+    print(
+        f"ResNet-18 GN ({n3/1e6:.2f}M params) batch {batch}: "
+        f"DP-SGD materialized {res[False]:.3f} ms | ghost {res[True]:.3f} ms "
+        f"({res[False] / res[True]:.2f}x faster, {batch * 1000 / res[True]:.0f} private samples/s)"
     )
 
 
