@@ -43,7 +43,9 @@ def test_grad_sample_linear_matches_autograd():
 
     torch.manual_seed(0)
     model = nn.Sequential(nn.Linear(6, 8), nn.ReLU(), nn.Linear(8, 3))
-    gsm = GradSampleModule(model)
+    # this test verifies the MATERIALIZED per-sample grads (ghost clipping
+    # intentionally never materializes them for single-token Linears)
+    gsm = GradSampleModule(model, ghost_clipping=False)
     x = torch.randn(4, 6)
     y = torch.randint(0, 3, (4,))
     loss = nn.functional.cross_entropy(gsm(x), y)
@@ -61,7 +63,8 @@ def test_grad_sample_conv_matches_autograd():
 
     torch.manual_seed(0)
     model = nn.Sequential(nn.Conv2d(2, 3, 3, padding=1), nn.ReLU(), nn.Flatten(), nn.Linear(3 * 4 * 4, 2))
-    gsm = GradSampleModule(model)
+    # materialized-path check (ghost clipping never builds grad_sample)
+    gsm = GradSampleModule(model, ghost_clipping=False)
     x = torch.randn(5, 2, 4, 4)
     y = torch.randint(0, 2, (5,))
     loss = nn.functional.cross_entropy(gsm(x), y, reduction="sum")
