@@ -346,3 +346,31 @@ def test_distributed_gather_path_matches_simulation(tmp_path):
     assert len(res["dist"]) == 2
     for (rd, ld), (rs, ls) in zip(res["dist"], res["sim"]):
         assert rd == rs and abs(ld - ls) < 5e-4, (res["dist"], res["sim"])
+
+
+def test_bench_script_two_rank_contract(tmp_path):
+    """bench.py's driver contract at N=2 (the SCALE run path): torchrun two
+    ranks over gloo, rank 0 prints exactly one JSON line with the required
+    fields and whole-job aggregate value."""
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29540",
+            "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+            "--local_steps", "1", "--batch_size", "16", "--shard_size", "128",
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-1500:]
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["scaling"] == "weak" and rec["higher_is_better"]
+    assert rec["steps"] == 2 and rec["warmup"] == 1
+    # whole-job aggregate: value * elapsed == world * local_steps * batch * steps
+    expected_samples = 2 * 1 * 16 * 2
+    elapsed_s = rec["ms_per_step"] * rec["steps"] / 1000.0
+    assert abs(rec["value"] * elapsed_s - expected_samples) / expected_samples < 1e-6
+    assert rec["config"]["global_batch"] == 32
