@@ -26,6 +26,8 @@ void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, 
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
 void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
                    float*, float, float, int64_t, int, int, int, int, hipStream_t);
+void launch_conv3x3_fwd(const void*, const void*, const float*, void*, int, int, int, int, int,
+                        hipStream_t);
 void launch_mkmmd_sums(const float*, const float*, double*, float*, int, int64_t, int64_t, int,
                        hipStream_t);
 void launch_mkmmd_backward(const float*, const float*, const float*, float*, int, int64_t,
@@ -266,6 +268,25 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tens
   return {dx, dgamma, dbeta};
 }
 
+// Direct 3x3 s1 p1 NHWC bf16 conv forward on MFMA (round-2 conv prototype).
+// x: [N, H, W, C] bf16 (channels-last); w: [9, C, K] bf16 prepacked
+// (torch weight [K, C, 3, 3] -> permute(2, 3, 1, 0).reshape(9, C, K));
+// returns [N, H, W, K] bf16.
+torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.dim() == 4 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 && w.dim() == 3 && w.is_contiguous());
+  TORCH_CHECK(w.size(0) == 9 && w.size(1) == x.size(3), "w must be [9, C, K]");
+  int64_t N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3), K = w.size(2);
+  TORCH_CHECK(W <= 32, "conv3x3_fwd prototype supports W <= 32");
+  if (bias.has_value()) check_f32(*bias, "bias");
+  auto y = torch::empty({N, H, W, K}, x.options());
+  launch_conv3x3_fwd(x.data_ptr(), w.data_ptr(),
+                     bias.has_value() ? bias->data_ptr<float>() : nullptr, y.data_ptr(),
+                     (int)N, (int)H, (int)W, (int)C, (int)K, stream());
+  return y;
+}
+
 // Fused multi-bandwidth Gaussian-kernel sums over a pairwise-distance Gram
 // (reference losses/mkmmd_loss.py:96-135).
 torch::Tensor mkmmd_sums(torch::Tensor d, torch::Tensor gammas, bool skip_diag) {
@@ -314,6 +335,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("clip_rowsum_", &clip_rowsum_, "clipped per-sample grad sum");
   m.def("confusion_counts_", &confusion_counts_, "streaming TP/FP/FN/TN counts");
   m.def("weighted_sum_rows", &weighted_sum_rows, "out = sum_k w[k]*stack[k]");
+  m.def("conv3x3_fwd", &conv3x3_fwd, "direct 3x3 NHWC bf16 conv forward (MFMA)");
   m.def("mkmmd_sums", &mkmmd_sums, "per-bandwidth Gaussian kernel sums over a Gram");
   m.def("mkmmd_backward", &mkmmd_backward, "dL/dGram for mkmmd_sums");
 }

@@ -24,6 +24,7 @@ setup(
                 os.path.join(CSRC, "flat_ops.hip"),
                 os.path.join(CSRC, "bn_ops.hip"),
                 os.path.join(CSRC, "mmd_ops.hip"),
+                os.path.join(CSRC, "conv_ops.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3"],

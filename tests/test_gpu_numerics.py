@@ -325,3 +325,38 @@ def test_mkmmd_sums_deterministic_gpu():
     a = F.mkmmd_sums(d, gammas, True)
     b = F.mkmmd_sums(d, gammas, True)
     assert torch.equal(a, b)
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [(8, 32, 32, 64, 64), (8, 8, 8, 256, 128), (8, 4, 4, 512, 64)])
+def test_conv3x3_fwd_mfma_matches_miopen(shape):
+    """Direct 3x3 MFMA conv prototype vs torch/MIOpen (bf16 tolerance)."""
+    from fl4health_amd import _C
+
+    torch.manual_seed(0)
+    n, h, w, c, k = shape
+    x = torch.randn(n, c, h, w, device="cuda", dtype=torch.bfloat16)
+    weight = torch.randn(k, c, 3, 3, device="cuda", dtype=torch.bfloat16) * 0.05
+    ref = torch.nn.functional.conv2d(x.contiguous(memory_format=torch.channels_last), weight, padding=1)
+    wp = weight.permute(2, 3, 1, 0).reshape(9, c, k).contiguous()
+    out = _C.conv3x3_fwd(x.permute(0, 2, 3, 1).contiguous(), wp, None).permute(0, 3, 1, 2)
+    rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max().clamp(min=1e-6)
+    assert rel < 2e-2, float(rel)
+
+
+@requires_gpu
+def test_conv3x3_fwd_bias_and_ragged_channels():
+    from fl4health_amd import _C
+
+    torch.manual_seed(1)
+    n, h, w, c, k = 4, 16, 16, 96, 80  # non-multiples of the 64-wide tiles
+    x = torch.randn(n, c, h, w, device="cuda", dtype=torch.bfloat16)
+    weight = torch.randn(k, c, 3, 3, device="cuda", dtype=torch.bfloat16) * 0.05
+    bias = torch.randn(k, device="cuda")
+    ref = torch.nn.functional.conv2d(
+        x.contiguous(memory_format=torch.channels_last), weight, bias.to(torch.bfloat16), padding=1
+    )
+    wp = weight.permute(2, 3, 1, 0).reshape(9, c, k).contiguous()
+    out = _C.conv3x3_fwd(x.permute(0, 2, 3, 1).contiguous(), wp, bias).permute(0, 3, 1, 2)
+    rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max().clamp(min=1e-6)
+    assert rel < 3e-2, float(rel)
