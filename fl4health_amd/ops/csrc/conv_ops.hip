@@ -69,27 +69,42 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kernel(
 
   for (int c0 = 0; c0 < C; c0 += CONV_CB) {
     int cb = min(CONV_CB, C - c0);
-    // ---- stage input halo tiles, one per packed sample (zero-padded) ----
-    for (int idx = tid; idx < sb * (bh + 2) * tile_w * CONV_CB; idx += CONV_THREADS) {
-      int c = idx % CONV_CB;
-      int pos = idx / CONV_CB;
-      int col = pos % tile_w;       // 0..W+1 -> image col col-1
-      int row = (pos / tile_w) % (bh + 2);  // 0..bh+1 -> image row h0+row-1
+    // ---- stage input halo tiles, 16 B per thread (8 channels) ----
+    for (int idx = tid; idx < sb * (bh + 2) * tile_w * (CONV_CB / 8); idx += CONV_THREADS) {
+      int cv = idx % (CONV_CB / 8);        // 8-channel vector index
+      int pos = idx / (CONV_CB / 8);
+      int col = pos % tile_w;              // 0..W+1 -> image col col-1
+      int row = (pos / tile_w) % (bh + 2); // 0..bh+1 -> image row h0+row-1
       int s = pos / (tile_w * (bh + 2));
       int ih = h0 + row - 1, iw = col - 1;
-      bf16 v = (bf16)0.0f;
-      if (c < cb && ih >= 0 && ih < H && iw >= 0 && iw < W)
-        v = x[(((int64_t)(n0 + s) * H + ih) * W + iw) * C + c0 + c];
-      s_in[pos * CONV_CB + c] = v;
+      bf16x8 v = bf16x8{};
+      int c = cv * 8;
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W && c + 8 <= cb)
+        v = *reinterpret_cast<const bf16x8*>(
+            &x[(((int64_t)(n0 + s) * H + ih) * W + iw) * C + c0 + c]);
+      else if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+        for (int j = 0; j < 8; ++j)
+          v[j] = (c + j < cb) ? x[(((int64_t)(n0 + s) * H + ih) * W + iw) * C + c0 + c + j]
+                              : (bf16)0.0f;
+      *reinterpret_cast<bf16x8*>(&s_in[pos * CONV_CB + c]) = v;
     }
-    // ---- stage weights for this c-chunk: [9][cb][KB] ----
-    for (int idx = tid; idx < 9 * CONV_CB * CONV_KB; idx += CONV_THREADS) {
-      int k = idx % CONV_KB;
-      int c = (idx / CONV_KB) % CONV_CB;
-      int tap = idx / (CONV_KB * CONV_CB);
-      bf16 v = (bf16)0.0f;
-      if (c < cb && kb0 + k < K) v = w[((int64_t)tap * C + c0 + c) * K + kb0 + k];
-      s_w[(tap * CONV_KB + k) * CONV_CB + c] = v;  // k-major: B frags contiguous
+    // ---- stage weights for this c-chunk, 16 B per thread along k ----
+    // global w is [9][C][K] (k contiguous); LDS is k-major [9][KB][CB] so the
+    // 8-wide global read scatters to 8 LDS rows (stride CB) — still 8x fewer
+    // global load instructions than scalar staging.
+    for (int idx = tid; idx < 9 * CONV_CB * (CONV_KB / 8); idx += CONV_THREADS) {
+      int kv = idx % (CONV_KB / 8);
+      int c = (idx / (CONV_KB / 8)) % CONV_CB;
+      int tap = idx / ((CONV_KB / 8) * CONV_CB);
+      int k = kv * 8;
+      bf16x8 v = bf16x8{};
+      if (c < cb && kb0 + k + 8 <= K)
+        v = *reinterpret_cast<const bf16x8*>(&w[((int64_t)tap * C + c0 + c) * K + kb0 + k]);
+      else if (c < cb)
+        for (int j = 0; j < 8; ++j)
+          v[j] = (kb0 + k + j < K) ? w[((int64_t)tap * C + c0 + c) * K + kb0 + k + j] : (bf16)0.0f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s_w[(tap * CONV_KB + k + j) * CONV_CB + c] = v[j];
     }
     __syncthreads();
 
