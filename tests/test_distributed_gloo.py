@@ -374,3 +374,37 @@ def test_bench_script_two_rank_contract(tmp_path):
     elapsed_s = rec["ms_per_step"] * rec["steps"] / 1000.0
     assert abs(rec["value"] * elapsed_s - expected_samples) / expected_samples < 1e-6
     assert rec["config"]["global_batch"] == 32
+
+
+WORKER_EVAL_FAULT = WORKER_FAULT.replace(
+    """    def fit(self, parameters, config):
+        if int(config["current_server_round"]) == 1 and self.client_name == "flaky":
+            raise RuntimeError("injected client fault")
+        return super().fit(parameters, config)""",
+    """    def evaluate(self, parameters, config):
+        if self.client_name == "flaky":
+            raise RuntimeError("injected evaluate fault")
+        return super().evaluate(parameters, config)""",
+)
+
+
+def test_distributed_evaluate_fault_containment(tmp_path):
+    """A client exception during the EVALUATE phase must not deadlock the
+    round: the server aggregates the surviving cohort's metrics."""
+    script = tmp_path / "worker_eval_fault.py"
+    script.write_text(WORKER_EVAL_FAULT)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29541",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    assert line, out.stdout[-2000:]
+    losses = json.loads(line[0][7:])["losses"]
+    assert len(losses) == 2  # both rounds complete; loss aggregated over rank 0 only
