@@ -360,3 +360,36 @@ def test_conv3x3_fwd_bias_and_ragged_channels():
     out = _C.conv3x3_fwd(x.permute(0, 2, 3, 1).contiguous(), wp, bias).permute(0, 3, 1, 2)
     rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max().clamp(min=1e-6)
     assert rel < 3e-2, float(rel)
+
+
+@requires_gpu
+def test_cdna_conv2d_module_fwd_bwd_matches_torch():
+    """CdnaConv2d full autograd on GPU: MFMA forward, rotated-weight MFMA
+    backward-data, shifted-GEMM backward-weights — vs nn.Conv2d autograd."""
+    from fl4health_amd.ops.conv import CdnaConv2d, convert_conv3x3_to_cdna
+
+    torch.manual_seed(0)
+    n, c, k, h, w = 8, 64, 64, 16, 16
+    ref = torch.nn.Conv2d(c, k, 3, padding=1).cuda()
+    ours = convert_conv3x3_to_cdna(torch.nn.Conv2d(c, k, 3, padding=1)).cuda()
+    assert isinstance(ours, CdnaConv2d)
+    ours.load_state_dict(ref.state_dict())
+
+    x1 = (torch.randn(n, c, h, w, device="cuda") * 0.5).contiguous(memory_format=torch.channels_last)
+    x1.requires_grad_(True)
+    x2 = x1.detach().clone().requires_grad_(True)
+
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        y_ref = ref(x1)
+    y_ours = ours(x2.to(torch.bfloat16))
+    assert y_ours.dtype == torch.bfloat16
+    rel = (y_ours.float() - y_ref.float()).abs().max() / y_ref.float().abs().max().clamp(min=1e-6)
+    assert rel < 3e-2, float(rel)
+
+    gy = torch.randn_like(y_ref)
+    y_ref.backward(gy)
+    y_ours.backward(gy.to(torch.bfloat16))
+    for name, (pr, po) in {"dx": (x1.grad, x2.grad), "dw": (ref.weight.grad, ours.weight.grad),
+                           "db": (ref.bias.grad, ours.bias.grad)}.items():
+        r = (po.float() - pr.float()).abs().max() / pr.float().abs().max().clamp(min=1e-6)
+        assert r < 5e-2, f"{name}: {float(r)}"

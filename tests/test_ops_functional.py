@@ -169,3 +169,56 @@ def test_weighted_sum_rows():
     w = torch.tensor([0.2, 0.3, 0.5])
     out = F.weighted_sum_rows(stack, w)
     assert torch.allclose(out, (w.unsqueeze(1) * stack).sum(0), atol=1e-6)
+
+
+def test_cdna_conv_packing_math_cpu():
+    """_pack_fwd/_pack_bwd layouts verified against torch conv semantics by
+    emulating the kernel contraction with einsum (the kernel itself is
+    hardware-validated in tests/test_gpu_numerics.py)."""
+    import torch.nn.functional as Fn
+
+    from fl4health_amd.ops.conv import _pack_bwd, _pack_fwd
+
+    torch.manual_seed(0)
+    n, c, k, h, w = 2, 5, 7, 6, 6
+    x = torch.randn(n, c, h, w)
+    weight = torch.randn(k, c, 3, 3)
+
+    def kernel_emulate(x_nchw, packed):
+        # packed: [9, Cin, Cout]; emulate y[n,h,w,o] = sum_tap,c xpad * packed
+        cin, cout = packed.shape[1], packed.shape[2]
+        xp = Fn.pad(x_nchw, (1, 1, 1, 1)).permute(0, 2, 3, 1)  # NHWC padded
+        y = torch.zeros(x_nchw.shape[0], h, w, cout)
+        for tap in range(9):
+            dy, dx = tap // 3, tap % 3
+            xs = xp[:, dy : dy + h, dx : dx + w, :]
+            y += torch.einsum("nhwc,co->nhwo", xs, packed[tap])
+        return y.permute(0, 3, 1, 2)
+
+    ref = Fn.conv2d(x, weight, padding=1)
+    got = kernel_emulate(x, _pack_fwd(weight))
+    assert torch.allclose(got, ref, atol=1e-4), (got - ref).abs().max()
+
+    # bwd_data: conv of gy with the flipped/transposed pack equals autograd dx
+    xg = x.clone().requires_grad_(True)
+    y2 = Fn.conv2d(xg, weight, padding=1)
+    gy = torch.randn_like(y2)
+    y2.backward(gy)
+    dx = kernel_emulate(gy, _pack_bwd(weight))
+    assert torch.allclose(dx, xg.grad, atol=1e-4), (dx - xg.grad).abs().max()
+
+
+def test_cdna_conv_class_swap_and_fallback_cpu():
+    from fl4health_amd.ops.conv import CdnaConv2d, convert_conv3x3_to_cdna
+
+    torch.manual_seed(0)
+    ref = torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3, padding=1), torch.nn.ReLU(),
+                              torch.nn.Conv2d(8, 8, 3, padding=1, bias=False))
+    model = convert_conv3x3_to_cdna(torch.nn.Sequential(
+        torch.nn.Conv2d(3, 8, 3, padding=1), torch.nn.ReLU(),
+        torch.nn.Conv2d(8, 8, 3, padding=1, bias=False)))
+    model.load_state_dict(ref.state_dict())  # state_dict compatible
+    assert isinstance(model[0], CdnaConv2d) and isinstance(model[2], CdnaConv2d)
+    x = torch.randn(2, 3, 8, 8)
+    # CPU path falls back to F.conv2d exactly
+    assert torch.allclose(model(x), ref(x), atol=1e-6)
