@@ -55,6 +55,11 @@ class BenchFedProxClient(FedProxClient):
 
             model = convert_batchnorm_to_cdna(model.to(memory_format=torch.channels_last))
             model = fuse_resnet_bn_relu(model)
+            if self.args.cdna_conv:
+                # opt-in: hand-written MFMA direct conv instead of MIOpen
+                from fl4health_amd.ops.conv import convert_conv3x3_to_cdna
+
+                model = convert_conv3x3_to_cdna(model)
         return model
 
     def get_data_loaders(self, config):
@@ -120,6 +125,8 @@ def main() -> None:
     parser.add_argument("--shard_size", type=int, default=8192)
     parser.add_argument("--no_graph", action="store_true", help="disable hipGraph train-step capture")
     parser.add_argument("--no_mirror", action="store_true", help="disable persistent bf16 weight mirrors")
+    parser.add_argument("--cdna_conv", action="store_true",
+                        help="use the hand-written MFMA direct 3x3 conv (CdnaConv2d) instead of MIOpen")
     args = parser.parse_args()
 
     set_all_random_seeds(42)
