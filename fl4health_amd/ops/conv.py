@@ -74,7 +74,14 @@ class _CdnaConv3x3Fn(torch.autograd.Function):
 class CdnaConv2d(nn.Conv2d):
     """Drop-in nn.Conv2d for 3x3/s1/p1/groups=1 that runs the MFMA direct
     kernel on GPU bf16 channels-last inputs; falls back to F.conv2d off that
-    fast path (CPU, fp32 eval, W > 32, other hyper-params)."""
+    fast path (CPU, fp32 eval, unsupported widths, other hyper-params).
+
+    `mfma_widths` gates adoption to the spatial widths where the kernel
+    MEASURES faster than MIOpen (profiles/kernels_summary.md: 16x16 1.75x,
+    8x8 1.76x; 32x32 and 4x4 still lose and fall back). Set it to
+    range(1, 33) to force the kernel everywhere (micro-benchmarks)."""
+
+    mfma_widths: frozenset = frozenset({8, 16})
 
     def _fast_path(self, input: torch.Tensor) -> bool:
         return (
@@ -86,7 +93,7 @@ class CdnaConv2d(nn.Conv2d):
             and self.padding == (1, 1)
             and self.dilation == (1, 1)
             and self.groups == 1
-            and input.shape[3] <= 32
+            and input.shape[3] in self.mfma_widths
         )
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:
