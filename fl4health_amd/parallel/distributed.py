@@ -231,6 +231,16 @@ class DistributedRuntime:
             total_examples = sum(g["n"] for g in ok.values())
             cohort_size = len(ok)
             layout = next(iter(ok.values()))
+            # every participating rank evaluates this on the SAME gathered
+            # metadata: a heterogeneous payload (client bug, or a strategy that
+            # should have forced the gather path) fails loudly and consistently
+            # on all ranks instead of hanging the size-mismatched all-reduce
+            if any(g["numels"] != layout["numels"] for g in ok.values()):
+                raise RuntimeError(
+                    "collective aggregation requires homogeneous client payloads; "
+                    f"got numels {[g['numels'] for g in ok.values()]} — use a strategy "
+                    "with supports_collective_aggregation() == False for per-client layouts"
+                )
             total_numel = sum(layout["numels"])
             # 2) pre-scaled all-reduce over ONE concatenated comm buffer
             if self.rank in ok and fit_res is not None and strategy is not None:

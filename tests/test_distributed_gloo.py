@@ -408,3 +408,42 @@ def test_distributed_evaluate_fault_containment(tmp_path):
     assert line, out.stdout[-2000:]
     losses = json.loads(line[0][7:])["losses"]
     assert len(losses) == 2  # both rounds complete; loss aggregated over rank 0 only
+
+
+WORKER_MISMATCH = WORKER_FAULT.replace(
+    """class FlakyClient(TinyClient):
+    # Fails fit on the FIRST round only (simulated transient client fault).
+
+    def fit(self, parameters, config):
+        if int(config["current_server_round"]) == 1 and self.client_name == "flaky":
+            raise RuntimeError("injected client fault")
+        return super().fit(parameters, config)""",
+    """class FlakyClient(TinyClient):
+    # Returns a MALFORMED payload (extra tensor) on one rank: the collective
+    # path must fail loudly and consistently, never hang the all-reduce.
+
+    def fit(self, parameters, config):
+        params, n, metrics = super().fit(parameters, config)
+        if self.client_name == "flaky":
+            params = type(params)(list(params.tensors) + [torch.zeros(3)], dict(params.meta))
+        return params, n, metrics""",
+)
+
+
+def test_distributed_heterogeneous_payload_fails_loud(tmp_path):
+    """A size-mismatched payload on the collective fast path must raise the
+    homogeneity error on every rank (exit != 0), not deadlock."""
+    script = tmp_path / "worker_mismatch.py"
+    script.write_text(WORKER_MISMATCH)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29542",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=300, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode != 0
+    assert "homogeneous client payloads" in (out.stderr + out.stdout)
