@@ -5,6 +5,8 @@ GradSampleModule/DpSgdEngine (Opacus-equivalent on CDNA4 kernels, K7).
 """
 from __future__ import annotations
 
+import secrets
+
 from fl4health_amd.clients.basic_client import BasicClient
 from fl4health_amd.common import Config
 from fl4health_amd.privacy.dp_sgd import DpSgdEngine
@@ -35,9 +37,18 @@ class InstanceLevelDpClient(BasicClient):
         gsm = GradSampleModule(self.model)
         self._gsm = gsm
         self.model = gsm
+        # DP noise MUST be unpredictable: a fixed/broadcast seed lets an
+        # observer regenerate and subtract the noise stream, voiding the DP
+        # guarantee. Default to a cryptographically random per-client per-run
+        # seed; a fixed seed is honored only via the explicit debug/test key
+        # `dp_noise_seed` (never the shared training `seed`).
+        if "dp_noise_seed" in config:
+            noise_seed = int(config["dp_noise_seed"])
+        else:
+            noise_seed = secrets.randbits(63)
         self.dp_engine = DpSgdEngine(
             gsm, self.optimizers["global"], self.noise_multiplier, self.clipping_bound,
-            seed=int(config.get("seed", 0)),
+            seed=noise_seed,
         )
 
     def set_optimizer_zero_grad(self) -> None:

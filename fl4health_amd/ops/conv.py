@@ -93,6 +93,7 @@ class CdnaConv2d(nn.Conv2d):
             and self.padding == (1, 1)
             and self.dilation == (1, 1)
             and self.groups == 1
+            and self.padding_mode == "zeros"  # kernel hardcodes zero halo
             and input.shape[3] in self.mfma_widths
         )
 
@@ -116,6 +117,7 @@ def convert_conv3x3_to_cdna(model: nn.Module) -> nn.Module:
             and m.padding == (1, 1)
             and m.dilation == (1, 1)
             and m.groups == 1
+            and m.padding_mode == "zeros"
         ):
             m.__class__ = CdnaConv2d
     return model

@@ -1,6 +1,6 @@
 // Python bindings for the CDNA4 flat-buffer FL kernels (see flat_ops.hip).
 #include <torch/extension.h>
-#include <ATen/cuda/CUDAContext.h>
+#include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
 extern "C" {
@@ -45,7 +45,7 @@ void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
 }
 
-hipStream_t stream() { return at::cuda::getCurrentCUDAStream().stream(); }
+hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
 
 void axpby_(torch::Tensor y, torch::Tensor x, double a, double b) {
   check_f32(y, "y");

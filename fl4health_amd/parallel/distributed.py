@@ -105,15 +105,32 @@ class DistributedRuntime:
         return out
 
     def _bcast_parameters(self, params: Parameters | None, src: int = 0) -> Parameters:
-        """Broadcast Parameters: one meta object + ONE concatenated tensor."""
+        """Broadcast Parameters: one meta object + ONE concatenated tensor.
+
+        The wire buffer is fp32, which is exact for every floating dtype up to
+        fp32 but silently corrupts int64 indices/masks above 2^24 — so integer
+        payloads ride in the header object instead of the fused buffer."""
         if self.rank == src:
             assert params is not None
-            numels = [int(t.numel()) for t in params.tensors]
-            shapes = [list(t.shape) for t in params.tensors]
-            header = {"numels": numels, "shapes": shapes, "meta": params.meta}
+            float_idx = [i for i, t in enumerate(params.tensors) if t.is_floating_point()]
+            int_payload = {
+                i: t.detach().cpu() for i, t in enumerate(params.tensors) if not t.is_floating_point()
+            }
+            numels = [int(params.tensors[i].numel()) for i in float_idx]
+            shapes = [list(params.tensors[i].shape) for i in float_idx]
+            dtypes = [str(params.tensors[i].dtype) for i in float_idx]
+            header = {
+                "n_tensors": len(params.tensors),
+                "float_idx": float_idx,
+                "numels": numels,
+                "shapes": shapes,
+                "dtypes": dtypes,
+                "int_payload": int_payload,
+                "meta": params.meta,
+            }
             self._bcast_obj(header, src=src)
             buf = (
-                torch.cat([t.reshape(-1).to(self.comm_device, torch.float32) for t in params.tensors])
+                torch.cat([params.tensors[i].reshape(-1).to(self.comm_device, torch.float32) for i in float_idx])
                 if numels
                 else torch.zeros(0, device=self.comm_device)
             )
@@ -123,11 +140,16 @@ class DistributedRuntime:
         total = sum(header["numels"])
         buf = torch.empty(total, dtype=torch.float32, device=self.comm_device)
         dist.broadcast(buf, src=src)
-        tensors = []
+        tensors: list[torch.Tensor | None] = [None] * header["n_tensors"]
         off = 0
-        for n, shp in zip(header["numels"], header["shapes"]):
-            tensors.append(buf[off : off + n].view(shp).clone())
+        for i, n, shp, dt in zip(header["float_idx"], header["numels"], header["shapes"], header["dtypes"]):
+            t = buf[off : off + n].view(shp).clone()
+            want = getattr(torch, dt.replace("torch.", ""))
+            tensors[i] = t.to(want) if want != torch.float32 else t
             off += n
+        for i, t in header["int_payload"].items():
+            tensors[i] = t.to(self.comm_device) if self.comm_device.type != "cpu" else t
+        assert all(t is not None for t in tensors)
         return Parameters(tensors, header["meta"])
 
     # ------------------------------------------------------------------

@@ -18,12 +18,27 @@ SUPPORTED_LAYERS = (nn.Linear, nn.Conv1d, nn.Conv2d, nn.GroupNorm, nn.LayerNorm,
 
 def validate_module(module: nn.Module) -> None:
     """BatchNorm mixes samples -> incompatible with per-sample DP (same rule
-    as Opacus; reference privacy_utilities.py)."""
+    as Opacus; reference privacy_utilities.py). Additionally, any trainable
+    parameter living on a module type we cannot hook (Conv3d, RNNs,
+    MultiheadAttention, custom containers with direct nn.Parameters...) would
+    silently keep its ordinary .grad and reach the optimizer UNCLIPPED and
+    UN-NOISED — a privacy leak, so it is a hard error (Opacus raises too).
+    Freeze such parameters (requires_grad=False) to exclude them explicitly."""
     for name, m in module.named_modules():
         if isinstance(m, (nn.BatchNorm1d, nn.BatchNorm2d, nn.BatchNorm3d)):
             raise ValueError(
                 f"module {name} is BatchNorm: incompatible with instance-level DP; "
                 "call convert_batchnorm_modules(model) first"
+            )
+        if isinstance(m, SUPPORTED_LAYERS):
+            continue
+        direct_trainable = [pn for pn, p in m.named_parameters(recurse=False) if p.requires_grad]
+        if direct_trainable:
+            raise ValueError(
+                f"module {name or '<root>'} ({type(m).__name__}) holds trainable parameters "
+                f"{direct_trainable} but is not a supported per-sample-grad layer "
+                f"({', '.join(t.__name__ for t in SUPPORTED_LAYERS)}); its gradients would "
+                "bypass DP clipping/noising. Freeze them or replace the module."
             )
 
 
@@ -118,13 +133,14 @@ class GradSampleModule(nn.Module):
             if m.bias is not None:
                 self._store(m.bias, g2.sum(dim=1))
         elif isinstance(m, (nn.Conv1d, nn.Conv2d)):
-            b = act.shape[0]
-            if (
-                self.ghost_clipping
-                and isinstance(m, nn.Conv2d)
-                and m.groups == 1
-                and m not in self._ghost
-            ):
+            if m in self._ghost:
+                # module fired more than once this step: the ghost-norm trick
+                # cannot see the cross-term between firings, so materialize the
+                # stored pair and fall through to the per-sample-grad path for
+                # both (mirrors the Linear fallback above)
+                a0, g0 = self._ghost.pop(m)
+                self._materialize_conv(m, a0, g0)
+            elif self.ghost_clipping and isinstance(m, nn.Conv2d) and m.groups == 1:
                 # ghost-norm pays when the [L, L] Grams are smaller than the
                 # per-sample grad itself: L^2 < |W| (deep ResNet blocks: L=64,
                 # |W|=590k). Large-spatial early convs stay materialized.
@@ -132,26 +148,7 @@ class GradSampleModule(nn.Module):
                 if l_spatial * l_spatial < m.weight.numel():
                     self._ghost[m] = (act, go)
                     return
-            if isinstance(m, nn.Conv2d):
-                unfolded = Fn.unfold(act, m.kernel_size, m.dilation, m.padding, m.stride)  # [B, Cin*k*k, L]
-            else:
-                # conv1d as a (1, k) conv2d for unfold
-                a4 = act.unsqueeze(2)  # [B, C, 1, L]
-                unfolded = Fn.unfold(
-                    a4, (1, m.kernel_size[0]), (1, m.dilation[0]), (0, m.padding[0]), (1, m.stride[0])
-                )  # [B, Cin*k, L]
-            g2 = go.reshape(b, go.shape[1], -1)  # [B, Cout, L]
-            if m.groups == 1:
-                gs = torch.einsum("bol,bil->boi", g2, unfolded)  # [B, Cout, Cin*k*k]
-            else:
-                cin_per_g = act.shape[1] // m.groups
-                cout_per_g = go.shape[1] // m.groups
-                uf = unfolded.reshape(b, m.groups, cin_per_g * int(torch.tensor(m.kernel_size).prod()), -1)
-                gg = g2.reshape(b, m.groups, cout_per_g, -1)
-                gs = torch.einsum("bgol,bgil->bgoi", gg, uf).reshape(b, go.shape[1], -1)
-            self._store(m.weight, gs.reshape(b, *m.weight.shape))
-            if m.bias is not None:
-                self._store(m.bias, g2.sum(dim=2))
+            self._materialize_conv(m, act, go)
         elif isinstance(m, (nn.GroupNorm, nn.LayerNorm)):
             if isinstance(m, nn.GroupNorm):
                 normed = Fn.group_norm(act, m.num_groups, eps=m.eps)
@@ -174,6 +171,30 @@ class GradSampleModule(nn.Module):
             idx = act.reshape(b, -1, 1).expand(-1, -1, m.embedding_dim)
             gs.scatter_add_(1, idx.long(), go.reshape(b, -1, m.embedding_dim))
             self._store(m.weight, gs)
+
+    def _materialize_conv(self, m: nn.Module, act: torch.Tensor, go: torch.Tensor) -> None:
+        """Explicit per-sample conv grads via unfold + batched einsum."""
+        b = act.shape[0]
+        if isinstance(m, nn.Conv2d):
+            unfolded = Fn.unfold(act, m.kernel_size, m.dilation, m.padding, m.stride)  # [B, Cin*k*k, L]
+        else:
+            # conv1d as a (1, k) conv2d for unfold
+            a4 = act.unsqueeze(2)  # [B, C, 1, L]
+            unfolded = Fn.unfold(
+                a4, (1, m.kernel_size[0]), (1, m.dilation[0]), (0, m.padding[0]), (1, m.stride[0])
+            )  # [B, Cin*k, L]
+        g2 = go.reshape(b, go.shape[1], -1)  # [B, Cout, L]
+        if m.groups == 1:
+            gs = torch.einsum("bol,bil->boi", g2, unfolded)  # [B, Cout, Cin*k*k]
+        else:
+            cin_per_g = act.shape[1] // m.groups
+            cout_per_g = go.shape[1] // m.groups
+            uf = unfolded.reshape(b, m.groups, cin_per_g * int(torch.tensor(m.kernel_size).prod()), -1)
+            gg = g2.reshape(b, m.groups, cout_per_g, -1)
+            gs = torch.einsum("bgol,bgil->bgoi", gg, uf).reshape(b, go.shape[1], -1)
+        self._store(m.weight, gs.reshape(b, *m.weight.shape))
+        if m.bias is not None:
+            self._store(m.bias, g2.sum(dim=2))
 
     # ------------------------------------------------------------------
     def clear_grad_samples(self) -> None:

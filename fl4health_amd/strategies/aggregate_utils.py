@@ -15,16 +15,25 @@ from fl4health_amd.ops import functional as F
 
 
 def pseudo_sort_key(cid: str, num_examples: int, parameters: Parameters) -> tuple:
-    """Deterministic ordering key (reference decode_and_pseudo_sort_results:
-    sorts by sample counts + tensor content signature to pin fp summation order)."""
-    return (num_examples, cid)
+    """Deterministic ordering key (reference utils/functions.py:63-82): sample
+    count + a content signature (sum of zeroeth elements of the floating
+    tensors) so that two clients with equal num_examples and unstable cids
+    still land in a pinned fp-summation order. The cid is kept as the final
+    tiebreak for the (measure-zero) case of identical signatures."""
+    sig = 0.0
+    for t in parameters.tensors:
+        if t.is_floating_point() and t.numel() > 0:
+            sig += float(t.reshape(-1)[0])
+    return (num_examples + sig, cid)
 
 
 def decode_and_pseudo_sort_results(
     results: list[tuple[object, FitRes]],
 ) -> list[tuple[object, Parameters, int]]:
     sortable = [(proxy, res.parameters, res.num_examples) for proxy, res in results]
-    return sorted(sortable, key=lambda t: (t[2], getattr(t[0], "cid", "")))
+    return sorted(
+        sortable, key=lambda t: pseudo_sort_key(str(getattr(t[0], "cid", "")), t[2], t[1])
+    )
 
 
 def aggregate_results(results: list[tuple[Parameters, int]], weighted: bool = True) -> Parameters:

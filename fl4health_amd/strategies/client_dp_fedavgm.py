@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import logging
 import math
+import secrets
 
 import torch
 
@@ -48,6 +49,7 @@ class ClientLevelDPFedAvgM(BasicFedAvg):
         clipping_noise_multiplier: float = 1.0,
         beta: float = 0.9,
         per_client_example_cap: float | None = None,
+        noise_seed: int | None = None,
         **kwargs,
     ) -> None:
         assert kwargs.get("initial_parameters") is not None, "initial parameters required"
@@ -67,7 +69,10 @@ class ClientLevelDPFedAvgM(BasicFedAvg):
         self.parameter_packer = ParameterPackerWithClippingBit()
         self.current_weights: torch.Tensor | None = None
         self.m_t: torch.Tensor | None = None
-        self._noise_seed = 1234
+        # Server-side DP noise must not come from a publicly-known constant
+        # (predictable noise is removable). Random per-run seed unless a test
+        # explicitly pins one via noise_seed.
+        self._noise_seed = noise_seed if noise_seed is not None else secrets.randbits(62)
 
     def add_auxiliary_information(self, original_parameters: Parameters) -> None:
         self.current_weights = original_parameters.tensors[0].detach().clone()

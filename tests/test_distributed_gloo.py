@@ -447,3 +447,52 @@ def test_distributed_heterogeneous_payload_fails_loud(tmp_path):
     )
     assert out.returncode != 0
     assert "homogeneous client payloads" in (out.stderr + out.stdout)
+
+
+BCAST_DTYPE_WORKER = r"""
+import torch
+from fl4health_amd.common import Parameters
+from fl4health_amd.parallel.distributed import DistributedRuntime
+
+t = DistributedRuntime(backend="gloo")
+if t.rank == 0:
+    params = Parameters(
+        [
+            torch.arange(5, dtype=torch.int64) + (1 << 40),  # > 2^24: dies in fp32
+            torch.full((3,), 1.5, dtype=torch.bfloat16),
+            torch.tensor([0.25, -0.5], dtype=torch.float32),
+            torch.tensor([1, 0, 1], dtype=torch.bool),
+        ],
+        {"tag": "dtype-roundtrip"},
+    )
+    out = t._bcast_parameters(params, src=0)
+else:
+    out = t._bcast_parameters(None, src=0)
+assert out.tensors[0].dtype == torch.int64, out.tensors[0].dtype
+assert int(out.tensors[0][4]) == (1 << 40) + 4
+assert out.tensors[1].dtype == torch.bfloat16
+assert float(out.tensors[1][0]) == 1.5
+assert out.tensors[2].dtype == torch.float32
+assert out.tensors[3].dtype == torch.bool and bool(out.tensors[3][2])
+assert out.meta["tag"] == "dtype-roundtrip"
+print("BCAST_DTYPE_OK rank", t.rank)
+"""
+
+
+def test_bcast_parameters_preserves_dtypes(tmp_path):
+    """Integer/bool/bf16 payloads must survive the broadcast exactly —
+    fp32-wire casting silently corrupts int64 above 2^24."""
+    script = tmp_path / "bcast_worker.py"
+    script.write_text(BCAST_DTYPE_WORKER)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29547",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=180, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert out.stdout.count("BCAST_DTYPE_OK") == 2

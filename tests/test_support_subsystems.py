@@ -406,3 +406,114 @@ def test_ghost_clipping_conv2d_matches_materialized():
         return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
 
     assert torch.allclose(run(True), run(False), atol=1e-5)
+
+
+def test_conv2d_ghost_repeat_fire_materializes():
+    """A ghost-eligible Conv2d firing twice in one step (weight sharing) must
+    fall back to materialization for BOTH firings — the Gram trick cannot see
+    the cross-term between the two contributions."""
+    from fl4health_amd.privacy.dp_sgd import DpSgdEngine
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    class TwiceConv(nn.Module):
+        def __init__(self):
+            super().__init__()
+            # 64x64x3x3 weight at 4x4 spatial: L^2 = 256 < |W| = 36864 -> ghost-eligible
+            self.conv = nn.Conv2d(64, 64, 3, padding=1)
+            self.head = nn.Linear(64 * 4 * 4, 3)
+
+        def forward(self, x):
+            h = torch.relu(self.conv(x))
+            h = self.conv(h)  # conv fires TWICE
+            return self.head(h.flatten(1))
+
+    def run(ghost):
+        set_all_random_seeds(7)
+        model = TwiceConv()
+        gsm = GradSampleModule(model, ghost_clipping=ghost)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        eng = DpSgdEngine(gsm, opt, noise_multiplier=0.0, clipping_bound=0.4, seed=2)
+        x = torch.randn(5, 64, 4, 4)
+        y = torch.randint(0, 3, (5,))
+        eng.zero_grad()
+        nn.functional.cross_entropy(gsm(x), y).backward()
+        eng.step()
+        return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+
+    assert torch.allclose(run(True), run(False), atol=1e-5)
+
+
+def test_grad_sample_rejects_unsupported_trainable_layers():
+    """Trainable params on un-hookable module types must be a hard error —
+    they would reach the optimizer unclipped/un-noised."""
+    import pytest
+
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    class WithRaw(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(4, 4)
+            self.scale = nn.Parameter(torch.ones(4))  # direct param on container
+
+        def forward(self, x):
+            return self.fc(x) * self.scale
+
+    with pytest.raises(ValueError, match="bypass DP"):
+        GradSampleModule(WithRaw())
+
+    with pytest.raises(ValueError, match="bypass DP"):
+        GradSampleModule(nn.Sequential(nn.Conv3d(2, 2, 3), nn.Flatten(), nn.Linear(2, 2)))
+
+    # freezing the offending params makes the model acceptable
+    m = WithRaw()
+    m.scale.requires_grad_(False)
+    GradSampleModule(m)
+
+
+def test_dp_noise_seed_is_random_by_default():
+    """DP noise must not come from a fixed public seed: two
+    engines built without an explicit dp_noise_seed draw different noise."""
+    from fl4health_amd.strategies.client_dp_fedavgm import ClientLevelDPFedAvgM
+    from fl4health_amd.common import Parameters
+
+    init = Parameters([torch.zeros(4)])
+    s1 = ClientLevelDPFedAvgM(initial_parameters=init)
+    s2 = ClientLevelDPFedAvgM(initial_parameters=init)
+    assert s1._noise_seed != s2._noise_seed  # 2^-62 collision probability
+    s3 = ClientLevelDPFedAvgM(initial_parameters=init, noise_seed=99)
+    assert s3._noise_seed == 99
+
+
+def test_cdna_conv_gate_rejects_nonzero_padding_mode():
+    """padding_mode != 'zeros' must fall back (kernel hardcodes zero halo)."""
+    from fl4health_amd.ops.conv import CdnaConv2d, convert_conv3x3_to_cdna
+
+    m = nn.Conv2d(8, 8, 3, padding=1, padding_mode="reflect")
+    net = nn.Sequential(m)
+    convert_conv3x3_to_cdna(net)
+    assert type(net[0]) is nn.Conv2d  # not swapped
+
+    c = CdnaConv2d(8, 8, 3, padding=1, padding_mode="reflect")
+    assert not c._fast_path(torch.zeros(1, 8, 16, 16))
+
+
+def test_pseudo_sort_content_signature_tiebreak():
+    """Equal num_examples + unstable cids: the content signature must still
+    pin the summation order (reference utils/functions.py:63-82)."""
+    from fl4health_amd.common import FitRes, Parameters
+    from fl4health_amd.strategies.aggregate_utils import decode_and_pseudo_sort_results
+
+    class P:
+        def __init__(self, cid):
+            self.cid = cid
+
+    t_small = Parameters([torch.full((3,), 1.0)])
+    t_big = Parameters([torch.full((3,), 2.0)])
+    r1 = [(P("zz"), FitRes(t_small, 10, {})), (P("aa"), FitRes(t_big, 10, {}))]
+    r2 = [(P("qq"), FitRes(t_big, 10, {})), (P("bb"), FitRes(t_small, 10, {}))]
+    s1 = decode_and_pseudo_sort_results(r1)
+    s2 = decode_and_pseudo_sort_results(r2)
+    # same content order regardless of cids / input order
+    assert torch.equal(s1[0][1].tensors[0], s2[0][1].tensors[0])
+    assert float(s1[0][1].tensors[0][0]) == 1.0
