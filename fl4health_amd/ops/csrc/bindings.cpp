@@ -26,6 +26,8 @@ void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, 
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
 void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
                    float*, float, float, int64_t, int, int, int, int, hipStream_t);
+void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int, int, int, int,
+                             int, hipStream_t);
 void launch_conv3x3_fwd(const void*, const void*, const float*, void*, int, int, int, int, int,
                         hipStream_t);
 void launch_mkmmd_sums(const float*, const float*, double*, float*, int, int64_t, int64_t, int,
@@ -287,6 +289,27 @@ torch::Tensor conv3x3_fwd(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+// Variant C of the direct conv: KB = 32, both operands glds-pipelined.
+// wimg: [K/32, C/64, 9, 32, 64] bf16 — the exact LDS image, packed host-side
+// (ops/conv.py pack_weight_kb32). Requires C % 64 == 0 and K % 32 == 0.
+torch::Tensor conv3x3_fwd_kb32(torch::Tensor x, torch::Tensor wimg,
+                               c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.dim() == 4 && x.is_contiguous());
+  TORCH_CHECK(wimg.is_cuda() && wimg.scalar_type() == torch::kBFloat16 && wimg.dim() == 5 &&
+              wimg.is_contiguous());
+  int64_t N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int64_t K = wimg.size(0) * 32;
+  TORCH_CHECK(wimg.size(1) * 64 == C && wimg.size(2) == 9 && wimg.size(3) == 32 &&
+              wimg.size(4) == 64, "wimg must be [K/32, C/64, 9, 32, 64]");
+  TORCH_CHECK(W <= 32, "conv3x3 prototype supports W <= 32");
+  if (bias.has_value()) check_f32(*bias, "bias");
+  auto y = torch::empty({N, H, W, K}, x.options());
+  launch_conv3x3_fwd_kb32(x.data_ptr(), wimg.data_ptr(),
+                          bias.has_value() ? bias->data_ptr<float>() : nullptr, y.data_ptr(),
+                          (int)N, (int)H, (int)W, (int)C, (int)K, stream());
+  return y;
+}
+
 // Fused multi-bandwidth Gaussian-kernel sums over a pairwise-distance Gram
 // (reference losses/mkmmd_loss.py:96-135).
 torch::Tensor mkmmd_sums(torch::Tensor d, torch::Tensor gammas, bool skip_diag) {
@@ -336,6 +359,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("confusion_counts_", &confusion_counts_, "streaming TP/FP/FN/TN counts");
   m.def("weighted_sum_rows", &weighted_sum_rows, "out = sum_k w[k]*stack[k]");
   m.def("conv3x3_fwd", &conv3x3_fwd, "direct 3x3 NHWC bf16 conv forward (MFMA)");
+  m.def("conv3x3_fwd_kb32", &conv3x3_fwd_kb32,
+        "direct 3x3 NHWC bf16 conv forward, KB=32 glds-pipelined variant");
   m.def("mkmmd_sums", &mkmmd_sums, "per-bandwidth Gaussian kernel sums over a Gram");
   m.def("mkmmd_backward", &mkmmd_backward, "dL/dGram for mkmmd_sums");
 }

@@ -22,10 +22,20 @@
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 #include <algorithm>
+#include <cstdlib>
 
 using bf16 = __bf16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// LDS bank-conflict swizzle (guide st_16x32): the halo/weight images are
+// 128-B rows (64 bf16 channels); a ds_read_b128 whose 16 lanes stride rows
+// 8-way-conflicts on the 64x4B banks. XOR channel bit 4 with row bit 2 to
+// spread the lane group over 4 bank slots. Applied BOTH sides: reads XOR the
+// LDS address; staging XORs (a) the ds_write channel for the plain path and
+// (b) the per-lane GLOBAL fetch address for glds (the glds LDS destination is
+// lane-linear and cannot be swizzled — guide rule 21).
+__device__ inline int conv_swz(int row) { return ((row >> 2) & 1) << 4; }
 
 #define CONV_KB 64      // output channels per block
 #define CONV_CB 64      // reduction channels per LDS stage
@@ -39,10 +49,11 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 __shared__ __bf16 s_conv[CONV_TILE_POS * CONV_CB + 9 * CONV_CB * CONV_KB];
 
 __device__ inline void stage_weights(const bf16* __restrict__ w, __bf16* __restrict__ s_w,
-                                     int C, int K, int c0, int cb, int kb0, int tid) {
+                                     int C, int K, int c0, int cb, int kb0, int tid,
+                                     int nthreads = CONV_THREADS) {
   // global w is [9][C][K] (k contiguous); LDS is k-major [9][KB][CB] so B
   // fragments (fixed out-channel, 8 consecutive c) are one 16-byte read.
-  for (int idx = tid; idx < 9 * CONV_CB * (CONV_KB / 8); idx += CONV_THREADS) {
+  for (int idx = tid; idx < 9 * CONV_CB * (CONV_KB / 8); idx += nthreads) {
     int kv = idx % (CONV_KB / 8);
     int c = (idx / (CONV_KB / 8)) % CONV_CB;
     int tap = idx / ((CONV_KB / 8) * CONV_CB);
@@ -54,7 +65,10 @@ __device__ inline void stage_weights(const bf16* __restrict__ w, __bf16* __restr
       for (int j = 0; j < 8; ++j)
         v[j] = (kb0 + k + j < K) ? w[((int64_t)tap * C + c0 + c) * K + kb0 + k + j] : (bf16)0.0f;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) s_w[(tap * CONV_KB + k + j) * CONV_CB + c] = v[j];
+    for (int j = 0; j < 8; ++j) {
+      int r = tap * CONV_KB + k + j;
+      s_w[r * CONV_CB + (c ^ conv_swz(r))] = v[j];
+    }
   }
 }
 
@@ -112,7 +126,7 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kernel(
           for (int j = 0; j < 8; ++j)
             v[j] = (c + j < cb) ? x[(((int64_t)(n0 + s) * H + ih) * W + iw) * C + c0 + c + j]
                                 : (bf16)0.0f;
-        *reinterpret_cast<bf16x8*>(&s_in[pos * CONV_CB + c]) = v;
+        *reinterpret_cast<bf16x8*>(&s_in[pos * CONV_CB + (c ^ conv_swz(pos))]) = v;
       }
       if (!persistent_w) stage_weights(w, s_w, C, K, c0, cb, kb0, tid);
       __syncthreads();
@@ -130,17 +144,18 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kernel(
             int m = wave * 32 + t * 16 + (lane & 15);
             int s = m / pps, rem = m % pps;
             int hh = rem / W, ww = rem % W;
-            const __bf16* src =
-                &s_in[((s * (bh + 2) + hh + dy) * tile_w + (ww + dx)) * CONV_CB + kbase];
+            int apos = (s * (bh + 2) + hh + dy) * tile_w + (ww + dx);
             // channel-contiguous: one 16-byte LDS read per fragment
-            bf16x8 a = *reinterpret_cast<const bf16x8*>(src);
+            bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                &s_in[apos * CONV_CB + (kbase ^ conv_swz(apos))]);
             if (m >= m_count) a = bf16x8{};
             afrag[t] = a;
           }
 #pragma unroll
           for (int q = 0; q < 4; ++q) {
-            const __bf16* bw = &s_w[(tap * CONV_KB + q * 16 + (lane & 15)) * CONV_CB + kbase];
-            bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(bw);
+            int br = tap * CONV_KB + q * 16 + (lane & 15);
+            bf16x8 bfrag =
+                *reinterpret_cast<const bf16x8*>(&s_w[br * CONV_CB + (kbase ^ conv_swz(br))]);
 #pragma unroll
             for (int t = 0; t < 2; ++t)
               acc[t][q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[t], bfrag, acc[t][q], 0, 0, 0);
@@ -172,19 +187,346 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Variant B: persistent-weight + glds double-buffered input (C == CONV_CB).
+//
+// The round-1 kernel at 32x32 C64 runs 1 block/CU with a serial
+// stage->sync->compute chain (0.55x MIOpen). Here the weights are staged once
+// per block, the grid is capped near the CU count, the block grid-strides over
+// M-tiles, and the NEXT tile's input halo is fetched with async
+// global_load_lds (16 B, lane-linear LDS image) WHILE the current tile's MFMAs
+// run; the single __syncthreads() per tile drains the DMA (guide §5: 2-buffer
+// glds + vmcnt(0) ties the best register pipeline in this 1-block/CU regime).
+// Out-of-range / halo lanes redirect their source address to a zero page —
+// the LDS destination of glds is wave-uniform base + lane*16 and cannot be
+// masked per-lane.
+// ---------------------------------------------------------------------------
+#define CONV_GLDS_CHUNK_CAP 1792   // 4-wave variant: 16 B chunks/buffer (32x32 M=128: 1632)
+#define CONV_GLDS_CHUNK_CAP8 2752  // 8-wave variant (32x32 M=256: 10*34*8 = 2720)
+// NOTE both caps are gated on round_up(chunks, 64): glds bases are 64-lane
+// aligned, so the ragged last group writes up to round_up(chunks, 64) - 1;
+// an exact-sized buffer lets it spill into the neighbouring buffer / the
+// persistent weights (the round-1 8-wave corruption bug).
+__device__ __align__(16) __bf16 conv_zero16[8] = {};
+
+__device__ inline void issue_glds_input(const bf16* __restrict__ x, __bf16* __restrict__ dst,
+                                        int n0, int h0, int sb, int bh, int H, int W, int C,
+                                        int c0, int tile_w, int tid, int nthreads) {
+  int wave = tid >> 6, lane = tid & 63;
+  int cpv = CONV_CB / 8;  // LDS image stride is CONV_CB channels per position
+  int total_chunks = sb * (bh + 2) * tile_w * cpv;
+  for (int base = wave * 64; base < total_chunks; base += nthreads) {
+    int idx = base + lane;
+    const bf16* src = conv_zero16;
+    if (idx < total_chunks) {
+      int cv = idx % cpv;
+      int pos = idx / cpv;
+      int col = pos % tile_w;
+      int row = (pos / tile_w) % (bh + 2);
+      int sidx = pos / (tile_w * (bh + 2));
+      int ih = h0 + row - 1, iw = col - 1;
+      int cf = (cv * 8) ^ conv_swz(pos);  // LDS slot cv*8 holds logical channel cf
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W && cf < C - c0)
+        src = &x[(((int64_t)(n0 + sidx) * H + ih) * W + iw) * C + c0 + cf];
+    }
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
+                                     (__attribute__((address_space(3))) void*)(dst + (size_t)base * 8),
+                                     16, 0, 0);
+  }
+}
+
+template <int WAVES>
+__global__ __launch_bounds__(WAVES * 64) void conv3x3_fwd_glds_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w, const float* __restrict__ bias,
+    bf16* __restrict__ y, int Nn, int H, int W, int C, int K, int BH, int SB, int n_tiles,
+    int h_groups) {
+  constexpr int NT = WAVES * 64;
+  constexpr int CAP = (WAVES == 8) ? CONV_GLDS_CHUNK_CAP8 : CONV_GLDS_CHUNK_CAP;
+  // 4-wave: 131 KB; 8-wave: 157 KB of the 160 KB LDS (1 WG/CU either way —
+  // the 8-wave variant exists to put 2 waves on each SIMD for latency hiding)
+  __shared__ __bf16 s_mem[2 * CAP * 8 + 9 * CONV_CB * CONV_KB];
+  __bf16* s_w = s_mem + 2 * CAP * 8;
+  int kb0 = blockIdx.z * CONV_KB;
+  int tile_w = W + 2;
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+
+  stage_weights(w, s_w, C, K, 0, C, kb0, tid, NT);  // persistent: staged ONCE
+  float breg[4];
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    int k = kb0 + q * 16 + (lane & 15);
+    breg[q] = (bias != nullptr && k < K) ? bias[k] : 0.0f;
+  }
+
+  auto tile_geom = [&](int tile, int& n0, int& h0, int& sb, int& bh) {
+    int hg = tile % h_groups;
+    n0 = (tile / h_groups) * SB;
+    sb = min(SB, Nn - n0);
+    h0 = hg * BH;
+    bh = min(BH, H - h0);
+  };
+
+  int tile = blockIdx.x;
+  if (tile < n_tiles) {
+    int n0, h0, sb, bh;
+    tile_geom(tile, n0, h0, sb, bh);
+    issue_glds_input(x, s_mem, n0, h0, sb, bh, H, W, C, 0, tile_w, tid, NT);
+  }
+  __syncthreads();  // drains prologue glds (vmcnt 0) + weight ds_writes
+  int cur = 0;
+
+  for (; tile < n_tiles; tile += gridDim.x) {
+    int n0, h0, sb, bh;
+    tile_geom(tile, n0, h0, sb, bh);
+    int pps = bh * W;
+    int m_count = sb * pps;
+    int nxt = tile + gridDim.x;
+    if (nxt < n_tiles) {
+      int nn0, nh0, nsb, nbh;
+      tile_geom(nxt, nn0, nh0, nsb, nbh);
+      issue_glds_input(x, s_mem + (1 - cur) * (CAP * 8), nn0, nh0, nsb, nbh, H, W, C, 0,
+                       tile_w, tid, NT);
+    }
+    const __bf16* s_in = s_mem + cur * (CAP * 8);
+
+    f32x4 acc[2][4];
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int q = 0; q < 4; ++q) acc[t][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      int dy = tap / 3, dx = tap % 3;
+#pragma unroll
+      for (int ck = 0; ck < CONV_CB / 32; ++ck) {
+        int kbase = ck * 32 + (lane >> 4) * 8;
+        bf16x8 afrag[2];
+#pragma unroll
+        for (int t = 0; t < 2; ++t) {
+          int m = wave * 32 + t * 16 + (lane & 15);
+          int sidx = m / pps, rem = m % pps;
+          int hh = rem / W, ww = rem % W;
+          int apos = (sidx * (bh + 2) + hh + dy) * tile_w + (ww + dx);
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &s_in[apos * CONV_CB + (kbase ^ conv_swz(apos))]);
+          if (m >= m_count) a = bf16x8{};
+          afrag[t] = a;
+        }
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          int br = tap * CONV_KB + q * 16 + (lane & 15);
+          bf16x8 bfrag =
+              *reinterpret_cast<const bf16x8*>(&s_w[br * CONV_CB + (kbase ^ conv_swz(br))]);
+#pragma unroll
+          for (int t = 0; t < 2; ++t)
+            acc[t][q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[t], bfrag, acc[t][q], 0, 0, 0);
+        }
+      }
+    }
+
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int m = wave * 32 + t * 16 + (lane >> 4) * 4 + r;
+          int k = q * 16 + (lane & 15);
+          if (m < m_count && kb0 + k < K) {
+            int sidx = m / pps, rem = m % pps;
+            int hh = rem / W, ww = rem % W;
+            y[(((int64_t)(n0 + sidx) * H + h0 + hh) * W + ww) * (int64_t)K + kb0 + k] =
+                (bf16)(acc[t][q][r] + breg[q]);
+          }
+        }
+    __syncthreads();  // all waves done reading s_in; prefetch glds drained
+    cur ^= 1;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Variant C: KB = 32, BOTH operands glds double-buffered, weights prepacked
+// host-side into the exact LDS image ([kz][cchunk][tap][kk 0..31][cc 0..63]
+// contiguous slabs) so weight staging is a pure lane-linear DMA — no address
+// math, no ds_write pass, and it overlaps the previous chunk's MFMAs. One
+// barrier per c-chunk. Built for the deep small-spatial shapes (4x4 C512:
+// round-1 kernel filled only 128 of 256 CUs; KB = 32 doubles kz and fills the
+// chip without split-C partials).
+// ---------------------------------------------------------------------------
+#define CONVC_KB 32
+#define CONVC_CHUNK_CAP 2304  // max input chunks (4x4 SB=8: 288 pos * 8)
+#define CONVC_WCHUNKS (9 * CONVC_KB * CONV_CB / 8)  // 2304 weight chunks/slab
+__shared__ __bf16 s_convc[2 * CONVC_CHUNK_CAP * 8 + 2 * CONVC_WCHUNKS * 8];
+
+__device__ inline void issue_glds_slab(const bf16* __restrict__ src_base, __bf16* __restrict__ dst,
+                                       int n_chunks, int tid) {
+  int wave = tid >> 6, lane = tid & 63;
+  for (int base = wave * 64; base < n_chunks; base += CONV_THREADS) {
+    const bf16* src = src_base + (size_t)(base + lane) * 8;  // slab is contiguous; cap is exact
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
+                                     (__attribute__((address_space(3))) void*)(dst + (size_t)base * 8),
+                                     16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kb32_kernel(
+    const bf16* __restrict__ x,
+    const bf16* __restrict__ wimg,  // [K/32][C/64][9][32][64] contiguous slabs
+    const float* __restrict__ bias, bf16* __restrict__ y, int Nn, int H, int W, int C, int K,
+    int BH, int SB, int n_tiles, int h_groups) {
+
+  int tile = blockIdx.x;
+  int kzi = blockIdx.z;
+  int kb0 = kzi * CONVC_KB;
+  int tile_w = W + 2;
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int n_cchunks = (C + CONV_CB - 1) / CONV_CB;
+
+  int hg = tile % h_groups;
+  int n0 = (tile / h_groups) * SB;
+  int sb = min(SB, Nn - n0);
+  int h0 = hg * BH;
+  int bh = min(BH, H - h0);
+  int pps = bh * W;
+  int m_count = sb * pps;
+
+  const bf16* wslab0 = wimg + ((size_t)kzi * n_cchunks) * (CONVC_WCHUNKS * 8);
+  issue_glds_input(x, s_convc, n0, h0, sb, bh, H, W, C, 0, tile_w, tid, CONV_THREADS);
+  issue_glds_slab(wslab0, s_convc + 2 * CONVC_CHUNK_CAP * 8, CONVC_WCHUNKS, tid);
+  __syncthreads();
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int q = 0; q < 2; ++q) acc[t][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  int cur = 0;
+  for (int ci = 0; ci < n_cchunks; ++ci) {
+    if (ci + 1 < n_cchunks) {
+      issue_glds_input(x, s_convc + (1 - cur) * (CONVC_CHUNK_CAP * 8), n0, h0, sb, bh, H, W, C,
+                       (ci + 1) * CONV_CB, tile_w, tid, CONV_THREADS);
+      issue_glds_slab(wslab0 + (size_t)(ci + 1) * (CONVC_WCHUNKS * 8),
+                      s_convc + 2 * CONVC_CHUNK_CAP * 8 + (1 - cur) * (CONVC_WCHUNKS * 8),
+                      CONVC_WCHUNKS, tid);
+    }
+    const __bf16* sin = s_convc + cur * (CONVC_CHUNK_CAP * 8);
+    const __bf16* sw = s_convc + 2 * CONVC_CHUNK_CAP * 8 + cur * (CONVC_WCHUNKS * 8);
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      int dy = tap / 3, dx = tap % 3;
+#pragma unroll
+      for (int ck = 0; ck < CONV_CB / 32; ++ck) {
+        int kbase = ck * 32 + (lane >> 4) * 8;
+        bf16x8 afrag[2];
+#pragma unroll
+        for (int t = 0; t < 2; ++t) {
+          int m = wave * 32 + t * 16 + (lane & 15);
+          int sidx = m / pps, rem = m % pps;
+          int hh = rem / W, ww = rem % W;
+          int apos = (sidx * (bh + 2) + hh + dy) * tile_w + (ww + dx);
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &sin[apos * CONV_CB + (kbase ^ conv_swz(apos))]);
+          if (m >= m_count) a = bf16x8{};
+          afrag[t] = a;
+        }
+#pragma unroll
+        for (int q = 0; q < 2; ++q) {
+          int br = tap * CONVC_KB + q * 16 + (lane & 15);
+          bf16x8 bfrag =
+              *reinterpret_cast<const bf16x8*>(&sw[br * CONV_CB + (kbase ^ conv_swz(br))]);
+#pragma unroll
+          for (int t = 0; t < 2; ++t)
+            acc[t][q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[t], bfrag, acc[t][q], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();  // drains next chunk's DMA; all waves done with cur
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int q = 0; q < 2; ++q)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int m = wave * 32 + t * 16 + (lane >> 4) * 4 + r;
+        int k = q * 16 + (lane & 15);
+        if (m < m_count && kb0 + k < K) {
+          int sidx = m / pps, rem = m % pps;
+          int hh = rem / W, ww = rem % W;
+          float v = acc[t][q][r];
+          if (bias != nullptr) v += bias[kb0 + k];
+          y[(((int64_t)(n0 + sidx) * H + h0 + hh) * W + ww) * (int64_t)K + kb0 + k] = (bf16)v;
+        }
+      }
+}
+
+static inline void conv_tile_geom(int Nn, int H, int W, int M, int& BH, int& SB, int& h_groups,
+                                  int& n_tiles) {
+  BH = std::min(std::max(M / W, 1), H);
+  SB = std::max(M / (H * W), 1);  // pack small images, several per block
+  SB = std::min(SB, Nn);
+  h_groups = (H + BH - 1) / BH;
+  n_tiles = ((Nn + SB - 1) / SB) * h_groups;
+}
+
 extern "C" void launch_conv3x3_fwd(const void* x, const void* w, const float* bias, void* y,
                                    int Nn, int H, int W, int C, int K, hipStream_t s) {
-  int BH = std::min(std::max(128 / W, 1), H);
-  int SB = std::max(128 / (H * W), 1);  // pack small images, several per block
-  SB = std::min(SB, Nn);
-  int h_groups = (H + BH - 1) / BH;
-  int n_tiles = ((Nn + SB - 1) / SB) * h_groups;
   int kz = (K + CONV_KB - 1) / CONV_KB;
-  // one tile per block: capping the grid to ~CU count and grid-striding was
-  // measured SLOWER on 32x32 (0.075 -> 0.089 ms) — serializing tiles costs
-  // more than the per-block weight stage saves at these sizes. The
-  // grid-stride machinery stays for the future glds-pipelined variant.
+  static int cap = [] {
+    const char* e = getenv("FL4_CONVB_GRID");
+    return e ? atoi(e) : 256;
+  }();
+  static int force_waves = [] {
+    const char* e = getenv("FL4_CONVB_WAVES");
+    return e ? atoi(e) : -1;  // -1 auto, 0 disable variant B, 4/8 force
+  }();
+  if (C <= CONV_CB && C % 8 == 0 && force_waves != 0) {
+    // Variant B: persistent weights + pipelined input DMA. Prefer the 8-wave
+    // (M = 256) instantiation — 2 waves/SIMD hides ds_read/MFMA latency at
+    // 1 block/CU — then the 4-wave one; fall through if the halo won't fit.
+    int BH, SB, h_groups, n_tiles;
+    conv_tile_geom(Nn, H, W, 256, BH, SB, h_groups, n_tiles);
+    int chunks8 = SB * (BH + 2) * (W + 2) * (CONV_CB / 8);
+    if (((chunks8 + 63) & ~63) <= CONV_GLDS_CHUNK_CAP8 && force_waves != 4) {
+      dim3 grid(std::min(n_tiles, cap), 1, kz);
+      conv3x3_fwd_glds_kernel<8><<<grid, 512, 0, s>>>(
+          (const bf16*)x, (const bf16*)w, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
+          h_groups);
+      return;
+    }
+    conv_tile_geom(Nn, H, W, 128, BH, SB, h_groups, n_tiles);
+    int chunks4 = SB * (BH + 2) * (W + 2) * (CONV_CB / 8);
+    if (((chunks4 + 63) & ~63) <= CONV_GLDS_CHUNK_CAP) {
+      dim3 grid(std::min(n_tiles, 2 * cap), 1, kz);
+      conv3x3_fwd_glds_kernel<4><<<grid, CONV_THREADS, 0, s>>>(
+          (const bf16*)x, (const bf16*)w, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
+          h_groups);
+      return;
+    }
+  }
+  int BH, SB, h_groups, n_tiles;
+  conv_tile_geom(Nn, H, W, 128, BH, SB, h_groups, n_tiles);
   dim3 grid(n_tiles, 1, kz);
   conv3x3_fwd_kernel<<<grid, CONV_THREADS, 0, s>>>(
       (const bf16*)x, (const bf16*)w, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles, h_groups);
+}
+
+// Variant C entry: caller supplies the LDS-image weight pack (see
+// ops/conv.py pack_weight_kb32). Requires C % 64 == 0, K % 32 == 0.
+extern "C" void launch_conv3x3_fwd_kb32(const void* x, const void* wimg, const float* bias,
+                                        void* y, int Nn, int H, int W, int C, int K,
+                                        hipStream_t s) {
+  int BH, SB, h_groups, n_tiles;
+  conv_tile_geom(Nn, H, W, 128, BH, SB, h_groups, n_tiles);
+  int kz = K / CONVC_KB;
+  dim3 grid(n_tiles, 1, kz);
+  conv3x3_fwd_kb32_kernel<<<grid, CONV_THREADS, 0, s>>>(
+      (const bf16*)x, (const bf16*)wimg, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
+      h_groups);
 }
