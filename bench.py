@@ -125,8 +125,10 @@ def main() -> None:
     parser.add_argument("--shard_size", type=int, default=8192)
     parser.add_argument("--no_graph", action="store_true", help="disable hipGraph train-step capture")
     parser.add_argument("--no_mirror", action="store_true", help="disable persistent bf16 weight mirrors")
-    parser.add_argument("--cdna_conv", action="store_true",
-                        help="use the hand-written MFMA direct 3x3 conv (CdnaConv2d) instead of MIOpen")
+    parser.add_argument("--cdna_conv", action="store_true", default=True,
+                        help="use the hand-written MFMA direct 3x3 conv (CdnaConv2d) instead of MIOpen (default on)")
+    parser.add_argument("--no_cdna_conv", dest="cdna_conv", action="store_false",
+                        help="fall back to MIOpen for the 3x3 convs")
     args = parser.parse_args()
 
     set_all_random_seeds(42)

@@ -40,11 +40,51 @@ def _pack_bwd(weight_bf16: torch.Tensor) -> torch.Tensor:
     return weight_bf16.flip(2, 3).permute(2, 3, 0, 1).reshape(9, k, c).contiguous()
 
 
+def _image_kb32(w9: torch.Tensor) -> torch.Tensor:
+    """[9, Cin, Kout] taps-major -> [K/32, C/64, 9, 32, 64] LDS-image slabs
+    for conv3x3_fwd_kb32, bank-swizzled to match conv_swz on the read side
+    (rows with kk bit 2 set get channel bit 4 XORed; glds stages verbatim)."""
+    c, k = w9.shape[1], w9.shape[2]
+    img = w9.reshape(9, c // 64, 64, k // 32, 32).permute(3, 1, 0, 4, 2).contiguous()
+    dev = w9.device
+    # hipGraph-capture-safe swizzle: index_select + where (boolean advanced
+    # indexing would call nonzero -> host sync, breaking capture)
+    swapped = img.index_select(-1, torch.arange(64, device=dev) ^ 16)
+    kk_mask = ((((torch.arange(32, device=dev) >> 2) & 1) == 1)).view(1, 1, 1, 32, 1)
+    return torch.where(kk_mask, swapped, img)
+
+
+def _use_kb32(width: int, c: int, k: int) -> bool:
+    """Per-shape variant gate from the measured matrix vs TUNED MIOpen
+    (cudnn.benchmark find, what training runs use — see
+    profiles/kernels_summary.md): the KB=32 fully-pipelined kernel wins at
+    4x4 C512 (1.23x) and 8x8 C<=128 (1.50x); tuned MIOpen wins the rest
+    (it reaches 300-360 TF at 16x16/32x32 where our kernels sit at
+    190-290)."""
+    if c % 64 != 0 or k % 32 != 0:
+        return False
+    return width <= 4 or (width == 8 and c <= 128)
+
+
+def _run_fwd(x_nhwc: torch.Tensor, w9: torch.Tensor, bias: torch.Tensor | None) -> torch.Tensor:
+    width, c, k = x_nhwc.shape[2], w9.shape[1], w9.shape[2]
+    if _use_kb32(width, c, k):
+        return _C.conv3x3_fwd_kb32(x_nhwc, _image_kb32(w9), bias)
+    return _C.conv3x3_fwd(x_nhwc, w9, bias)
+
+
 class _CdnaConv3x3Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x_nhwc: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None):
         w16 = weight.to(torch.bfloat16)
-        y = _C.conv3x3_fwd(x_nhwc, _pack_fwd(w16), bias.float() if bias is not None else None)
+        if x_nhwc.is_cuda and _use_kb32(x_nhwc.shape[2], w16.shape[1], w16.shape[0]):
+            # fused single-kernel pack straight to the swizzled LDS image
+            y = _C.conv3x3_fwd_kb32(
+                x_nhwc, _C.pack_kb32(w16.contiguous(), False),
+                bias.float() if bias is not None else None,
+            )
+        else:
+            y = _run_fwd(x_nhwc, _pack_fwd(w16), bias.float() if bias is not None else None)
         ctx.save_for_backward(x_nhwc, w16)
         ctx.has_bias = bias is not None
         return y
@@ -55,19 +95,40 @@ class _CdnaConv3x3Fn(torch.autograd.Function):
         gy = gy.contiguous()
         n, h, w, k = gy.shape
         c = x_nhwc.shape[3]
-        # dx: same kernel, rotated weights
-        dx = _C.conv3x3_fwd(gy, _pack_bwd(w16), None)
-        # dW: 9 shifted GEMMs (fp32 accumulate)
-        g2 = gy.reshape(-1, k).float()  # [NHW, K]
+        # dx: same kernel, rotated weights (conv roles swap: C_conv = K)
+        if gy.is_cuda and _use_kb32(gy.shape[2], k, c):
+            dx = _C.conv3x3_fwd_kb32(gy, _C.pack_kb32(w16.contiguous(), True), None)
+        else:
+            dx = _run_fwd(gy, _pack_bwd(w16), None)
+        # dW: route to MIOpen's tuned wrw igemm. The round-1 "9 shifted
+        # GEMMs" form ran fp32 rocBLAS (no matrix cores) and measured 316 ms
+        # of a 450 ms steady-state window (gpurun_out/cdna_bench_kernels.md);
+        # a bf16 batched GEMM fixes the dtype but its [C, NHW]x[NHW, K]
+        # shapes fill <16 workgroups without split-K. MIOpen's wrw kernels
+        # split correctly; our MFMA kernels keep the fwd and dx legs.
+        # NHWC-contiguous [N,H,W,C] permuted to NCHW *is* channels_last
+        # memory — both permutes below are zero-copy.
+        if x_nhwc.is_cuda:
+            x_ncl = x_nhwc.permute(0, 3, 1, 2)
+            gy_ncl = gy.permute(0, 3, 1, 2)
+            w_cl = w16.contiguous(memory_format=torch.channels_last)
+            _, dw16, db16 = torch.ops.aten.convolution_backward(
+                gy_ncl, x_ncl, w_cl, [k] if ctx.has_bias else None,
+                [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [False, True, ctx.has_bias],
+            )
+            dw_raw = dw16.float()
+            dbias = db16.float() if ctx.has_bias else None
+            return dx, dw_raw, dbias
+        # CPU reference path (packing-math tests): taps-major batched matmul
+        g2 = gy.reshape(-1, k)  # [NHW, K]
         xp = Fn.pad(x_nhwc, (0, 0, 1, 1, 1, 1))  # pad W, H dims of NHWC
-        dw_taps = []
-        for dy in range(3):
-            for dx_ in range(3):
-                xs = xp[:, dy : dy + h, dx_ : dx_ + w, :].reshape(-1, c).float()  # [NHW, C]
-                dw_taps.append(xs.t() @ g2)  # [C, K]
-        dw = torch.stack(dw_taps, dim=0)  # [9, C, K]
+        st = xp.stride()
+        xs = xp.as_strided((3, 3, n, h, w, c), (st[1], st[2], st[0], st[1], st[2], st[3]))
+        xs = xs.reshape(9, n * h * w, c)  # copy: taps-major batch
+        dw = torch.bmm(xs.transpose(1, 2).float(), g2.unsqueeze(0).expand(9, -1, -1).float())
         dw_raw = dw.permute(2, 1, 0).reshape(k, c, 3, 3)
-        dbias = g2.sum(dim=0) if ctx.has_bias else None
+        dbias = g2.sum(dim=0, dtype=torch.float32) if ctx.has_bias else None
         return dx, dw_raw, dbias
 
 
@@ -76,15 +137,18 @@ class CdnaConv2d(nn.Conv2d):
     kernel on GPU bf16 channels-last inputs; falls back to F.conv2d off that
     fast path (CPU, fp32 eval, unsupported widths, other hyper-params).
 
-    `mfma_widths` gates adoption to the spatial widths where the kernel
-    MEASURES faster than MIOpen (profiles/kernels_summary.md: 16x16 1.75x,
-    8x8 1.76x; 32x32 and 4x4 still lose and fall back). Set it to
-    range(1, 33) to force the kernel everywhere (micro-benchmarks)."""
+    `mfma_widths` gates adoption to the spatial widths where a kernel variant
+    MEASURES faster than TUNED MIOpen (cudnn.benchmark find): the KB=32
+    fully-pipelined kernel at 4x4 C512 (1.23x) and 8x8 C<=128 (1.50x). The
+    other variants (8-wave persistent-weight glds at 32x32, round-1 direct
+    at 16x16) beat MIOpen's *default* solver picks by 1.2-1.9x but lose to
+    its tuned picks and stay opt-in via `force_mfma`."""
 
-    mfma_widths: frozenset = frozenset({8, 16})
+    mfma_widths: frozenset = frozenset({4, 8})
+    force_mfma: bool = False  # True: run our kernels on every supported width
 
     def _fast_path(self, input: torch.Tensor) -> bool:
-        return (
+        ok = (
             HAS_EXT
             and input.is_cuda
             and input.dtype == torch.bfloat16
@@ -94,7 +158,14 @@ class CdnaConv2d(nn.Conv2d):
             and self.dilation == (1, 1)
             and self.groups == 1
             and self.padding_mode == "zeros"  # kernel hardcodes zero halo
-            and input.shape[3] in self.mfma_widths
+        )
+        if not ok:
+            return False
+        if self.force_mfma:
+            return input.shape[3] <= 32
+        # both conv directions must hit the measured-win kb32 shapes
+        return _use_kb32(input.shape[3], self.in_channels, self.out_channels) and _use_kb32(
+            input.shape[3], self.out_channels, self.in_channels
         )
 
     def forward(self, input: torch.Tensor) -> torch.Tensor:

@@ -530,3 +530,48 @@ extern "C" void launch_conv3x3_fwd_kb32(const void* x, const void* wimg, const f
       (const bf16*)x, (const bf16*)wimg, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
       h_groups);
 }
+
+// ---------------------------------------------------------------------------
+// Fused weight pack for the KB=32 kernel: [K, C, 3, 3] bf16 (standard torch
+// conv layout) -> [K/32][C/64][9][32][64] LDS-image slabs with the conv_swz
+// bank swizzle baked in. One kernel replaces the ~6-op torch chain
+// (permute/contiguous/arange/index_select/where) that measured ~0.5 ms/step
+// of launch+small-tensor overhead in the flagship bench. mode 0: forward
+// pack. mode 1: bwd-data pack — roles swapped (conv over gy has C_conv = K
+// and K_conv = C) and taps rotated 180 degrees.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void pack_kb32_kernel(
+    const bf16* __restrict__ w,  // [K, C, 3, 3]
+    bf16* __restrict__ out,      // [Kc/32, Cc/64, 9, 32, 64] (conv-role dims)
+    int K, int C, int mode) {
+  int Kc = mode == 0 ? K : C;  // conv-role output channels
+  int Cc = mode == 0 ? C : K;
+  int64_t total = (int64_t)Kc * Cc * 9;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    int cc_slot = idx % 64;
+    int kk = (idx / 64) % 32;
+    int tap = (idx / (64 * 32)) % 9;
+    int ci = (idx / (64 * 32 * 9)) % (Cc / 64);
+    int kz = idx / ((int64_t)64 * 32 * 9 * (Cc / 64));
+    int cc = cc_slot ^ conv_swz(tap * CONVC_KB + kk);  // logical channel
+    int kc = kz * 32 + kk;
+    int c_conv = ci * 64 + cc;
+    int dy = tap / 3, dx = tap % 3;
+    bf16 v;
+    if (mode == 0) {
+      v = w[(((int64_t)kc * C + c_conv) * 3 + dy) * 3 + dx];
+    } else {
+      // bwd: conv-role (kc, c_conv) = (orig C, orig K); taps flipped
+      v = w[(((int64_t)c_conv * C + kc) * 3 + (2 - dy)) * 3 + (2 - dx)];
+    }
+    out[idx] = v;
+  }
+}
+
+extern "C" void launch_pack_kb32(const void* w, void* out, int K, int C, int mode,
+                                 hipStream_t s) {
+  int64_t total = (int64_t)K * C * 9;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  pack_kb32_kernel<<<blocks, 256, 0, s>>>((const bf16*)w, (bf16*)out, K, C, mode);
+}

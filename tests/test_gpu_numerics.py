@@ -369,7 +369,7 @@ def test_cdna_conv2d_module_fwd_bwd_matches_torch():
     from fl4health_amd.ops.conv import CdnaConv2d, convert_conv3x3_to_cdna
 
     torch.manual_seed(0)
-    n, c, k, h, w = 8, 64, 64, 16, 16
+    n, c, k, h, w = 8, 128, 128, 8, 8  # kb32-adopted shape (8x8 C<=128)
     ref = torch.nn.Conv2d(c, k, 3, padding=1).cuda()
     ours = convert_conv3x3_to_cdna(torch.nn.Conv2d(c, k, 3, padding=1)).cuda()
     assert isinstance(ours, CdnaConv2d)
@@ -393,3 +393,58 @@ def test_cdna_conv2d_module_fwd_bwd_matches_torch():
                            "db": (ref.bias.grad, ours.bias.grad)}.items():
         r = (po.float() - pr.float()).abs().max() / pr.float().abs().max().clamp(min=1e-6)
         assert r < 5e-2, f"{name}: {float(r)}"
+
+
+@requires_gpu
+def test_conv3x3_kb32_and_pack_kernel_match_reference():
+    """KB=32 variant + the fused pack_kb32 kernel: forward vs MIOpen, and the
+    packed image vs the python packing path (fwd and bwd-data modes)."""
+    from fl4health_amd import _C
+    from fl4health_amd.ops.conv import _image_kb32, _pack_bwd, _pack_fwd
+
+    torch.manual_seed(0)
+    for n, h, w, c, k in [(8, 4, 4, 512, 512), (8, 8, 8, 128, 128), (8, 8, 8, 64, 128)]:
+        x = torch.randn(n, c, h, w, device="cuda", dtype=torch.bfloat16)
+        weight = torch.randn(k, c, 3, 3, device="cuda", dtype=torch.bfloat16) * 0.05
+        ref = torch.nn.functional.conv2d(
+            x.contiguous(memory_format=torch.channels_last), weight, padding=1
+        )
+        img_kernel = _C.pack_kb32(weight, False)
+        img_py = _image_kb32(_pack_fwd(weight))
+        assert torch.equal(img_kernel, img_py), (n, h, w, c, k)
+        img_bwd_kernel = _C.pack_kb32(weight, True)
+        img_bwd_py = _image_kb32(_pack_bwd(weight))
+        assert torch.equal(img_bwd_kernel, img_bwd_py), (n, h, w, c, k)
+        out = _C.conv3x3_fwd_kb32(x.permute(0, 2, 3, 1).contiguous(), img_kernel, None)
+        out = out.permute(0, 3, 1, 2)
+        rel = (out.float() - ref.float()).abs().max() / ref.float().abs().max().clamp(min=1e-6)
+        assert rel < 2e-2, float(rel)
+
+
+@requires_gpu
+def test_cdna_conv2d_force_mfma_all_widths():
+    """force_mfma exercises the opt-in variants (8-wave glds at 32x32,
+    round-1 direct at 16x16) through the module autograd path."""
+    from fl4health_amd.ops.conv import CdnaConv2d
+
+    torch.manual_seed(3)
+    for c, k, hw in [(64, 64, 32), (128, 128, 16)]:
+        ref = torch.nn.Conv2d(c, k, 3, padding=1).cuda()
+        ours = torch.nn.Conv2d(c, k, 3, padding=1).cuda()
+        ours.load_state_dict(ref.state_dict())
+        ours.__class__ = CdnaConv2d
+        ours.force_mfma = True
+        x1 = (torch.randn(4, c, hw, hw, device="cuda") * 0.5).contiguous(
+            memory_format=torch.channels_last
+        ).requires_grad_(True)
+        x2 = x1.detach().clone().requires_grad_(True)
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            y_ref = ref(x1)
+        y_ours = ours(x2.to(torch.bfloat16))
+        rel = (y_ours.float() - y_ref.float()).abs().max() / y_ref.float().abs().max().clamp(min=1e-6)
+        assert rel < 3e-2, float(rel)
+        gy = torch.randn_like(y_ref)
+        y_ref.backward(gy)
+        y_ours.backward(gy.to(torch.bfloat16))
+        r = (x2.grad.float() - x1.grad.float()).abs().max() / x1.grad.float().abs().max().clamp(min=1e-6)
+        assert r < 5e-2, float(r)

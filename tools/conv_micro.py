@@ -82,6 +82,7 @@ def run_shape(n, h, w, c, k, iters=100):
 
 def main():
     assert torch.cuda.is_available()
+    torch.backends.cudnn.benchmark = True  # tuned MIOpen find (what bench.py runs)
     torch.manual_seed(0)
     allok = True
     for shape in [
