@@ -496,3 +496,148 @@ def test_bcast_parameters_preserves_dtypes(tmp_path):
     )
     assert out.returncode == 0, out.stdout + out.stderr
     assert out.stdout.count("BCAST_DTYPE_OK") == 2
+
+
+CHAOS8_WORKER = r"""
+import json, logging, torch
+logging.disable(logging.ERROR)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.common import Parameters
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed
+from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+from tests.test_utils import TinyClient, TinyNet
+
+set_all_random_seeds(42)
+
+class ChaosClient(TinyClient):
+    # Rank 3 raises during fit in round 2; rank 5 raises during evaluate in
+    # round 3 - both must be contained (accept_failures) while the OTHER
+    # ranks' rounds complete through the device-tensor fast path.
+    def __init__(self, rank, **kw):
+        super().__init__(**kw)
+        self._rank = rank
+
+    def fit(self, parameters, config):
+        if self._rank == 3 and config.get("current_server_round") == 2:
+            raise RuntimeError("chaos: fit blows up on rank 3, round 2")
+        return super().fit(parameters, config)
+
+    def evaluate(self, parameters, config):
+        if self._rank == 5 and config.get("current_server_round") == 3:
+            raise RuntimeError("chaos: evaluate blows up on rank 5, round 3")
+        return super().evaluate(parameters, config)
+
+def strategy_factory():
+    init = Parameters([FlatParameterView(TinyNet()).flat.clone()])
+    return BasicFedAvg(
+        initial_parameters=init,
+        fraction_fit=1.0, fraction_evaluate=1.0,
+        min_fit_clients=2, min_evaluate_clients=2, min_available_clients=8,
+        accept_failures=True,
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 2},
+        on_evaluate_config_fn=lambda r: {"current_server_round": r},
+    )
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 4, "batch_size": 16}, strategy_factory())
+
+def client_factory(rank, world):
+    return ChaosClient(rank, seed=rank, metrics=[Accuracy()], device="cpu")
+
+hist = run_distributed(server_factory, client_factory, num_rounds=4,
+                       strategy_factory=strategy_factory, backend="gloo")
+import torch.distributed as dist
+if hist is not None:
+    print("RESULT " + json.dumps({"losses": hist.losses_distributed,
+                                  "n_rounds": len(hist.losses_distributed)}))
+"""
+
+
+def test_distributed_chaos_8_ranks(tmp_path):
+    """8-rank gloo matrix with chaos injection: a fit failure (rank 3, round
+    2) and an evaluate failure (rank 5, round 3) are contained while the
+    steady-state device-tensor protocol keeps serving the healthy ranks."""
+    script = tmp_path / "chaos8.py"
+    script.write_text(CHAOS8_WORKER)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "8",
+            "--master-addr", "127.0.0.1", "--master-port", "29549",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
+    line = [ln for ln in out.stdout.splitlines() if ln.startswith("RESULT ")]
+    assert line, out.stdout[-2000:]
+    res = json.loads(line[0][len("RESULT "):])
+    assert res["n_rounds"] == 4  # every round completed despite both failures
+
+
+FAST_PATH_WORKER = r"""
+import logging, torch
+logging.disable(logging.ERROR)
+from fl4health_amd.utils.random import set_all_random_seeds
+from fl4health_amd.client_managers.base import SimpleClientManager
+from fl4health_amd.common import Parameters
+from fl4health_amd.metrics.metrics import Accuracy
+from fl4health_amd.parameter_exchange.flat import FlatParameterView
+from fl4health_amd.servers.base_server import FlServer
+from fl4health_amd.simulation import run_distributed
+from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
+from tests.test_utils import TinyClient, TinyNet
+import fl4health_amd.simulation as sim
+
+set_all_random_seeds(7)
+
+def strategy_factory():
+    init = Parameters([FlatParameterView(TinyNet()).flat.clone()])
+    return BasicFedAvg(
+        initial_parameters=init,
+        on_fit_config_fn=lambda r: {"current_server_round": r, "local_steps": 2})
+
+def server_factory():
+    return FlServer(SimpleClientManager(), {"n_server_rounds": 4, "batch_size": 16}, strategy_factory())
+
+def client_factory(rank, world):
+    return TinyClient(seed=rank, metrics=[Accuracy()], device="cpu")
+
+runtime_holder = {}
+orig = sim.DistributedRuntime
+class Spy(orig):
+    def __init__(self, *a, **kw):
+        super().__init__(*a, **kw)
+        runtime_holder["rt"] = self
+sim.DistributedRuntime = Spy
+hist = run_distributed(server_factory, client_factory, num_rounds=4,
+                       strategy_factory=strategy_factory, backend="gloo")
+rt = runtime_holder["rt"]
+# round 1 learns the schema (object path); rounds 2-4 must ride device tensors
+assert rt._fast_fit_rounds >= 3, rt._fast_fit_rounds
+print("FASTPATH_OK", rt._fast_fit_rounds, "rank", rt.rank)
+"""
+
+
+def test_distributed_fit_fast_path_engages(tmp_path):
+    """After the schema-learning first round, fit rounds must run on the
+    device-tensor protocol (no object collectives on the steady path)."""
+    script = tmp_path / "fastpath.py"
+    script.write_text(FAST_PATH_WORKER)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29551",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=300, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
+    assert out.stdout.count("FASTPATH_OK") == 2
