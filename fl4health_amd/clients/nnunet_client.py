@@ -1,114 +1,362 @@
-"""Segmentation ("nnU-Net style") client.
+"""nnU-Net segmentation client (reference fl4health/clients/nnunet_client.py:71-935).
 
-Capability of reference fl4health/clients/nnunet_client.py:71-935 without the
-nnunetv2 dependency (not installed offline): plans generation from the local
-dataset (patch size / channels / classes as a JSON blob shipped through
-config bytes), deep-supervision loss handling, PolyLR schedule, gradient
-clipping, FedBN-compatible norm exclusion. If the real `nnunetv2` package is
-importable it can be slotted into get_model/get_data_loaders by the user.
+Full protocol depth, MI355X-native:
+
+- dataset **fingerprint** extraction from the local volumes
+  (preprocessing/nnunet.extract_fingerprint; reference :522-552)
+- **plans election**: when the server has no plans, this client plans the
+  experiment from its own data and returns the pickled plans through
+  ``get_properties`` (reference :826-885)
+- **plans localisation**: server plans are modified for the local dataset —
+  plans/dataset names, medians, per-config data identifiers, the 5%-of-voxels
+  batch cap (preprocessing/nnunet.create_local_plans; reference :388-495)
+- **preprocessing**: crop -> resample-to-target-spacing -> fg-percentile clip
+  + z-score per channel (reference maybe_preprocess :488-520)
+- **async multiprocess loading** with explicit child shutdown
+  (utils/nnunet_utils.AsyncPatchLoader + NnUNetDataLoaderWrapper; reference
+  :883-913)
+- **deep supervision**: model emits a pyramid; preds/targets travel as dicts
+  keyed by resolution and are re-listed for the loss (reference :625-743)
+- ignore-label masking for metrics (reference mask_data :705-758), PolyLR by
+  step, gradient clipping at 12.0, gc freeze after round 1 (reference :913-935)
+
+If the real ``nnunetv2`` package is importable, ``NnunetClient`` uses its
+ExperimentPlanner for plans election and accepts its plans dicts unchanged
+(they share the schema); everything else already speaks that schema.
 """
 from __future__ import annotations
 
-import json
+import gc
+import importlib.util
 import logging
+import pickle
 
 import torch
-from torch.utils.data import DataLoader, TensorDataset
+from torch.utils.data import DataLoader
 
 from fl4health_amd.clients.basic_client import BasicClient, TorchPredType, TorchTargetType
 from fl4health_amd.common import Config
-from fl4health_amd.models.unet3d import DeepSupervisionLoss, PolyLRScheduler, UNet3D
+from fl4health_amd.metrics.metric_managers import MetricManager
+from fl4health_amd.models.unet3d import DeepSupervisionLoss, UNet3D
+from fl4health_amd.preprocessing.nnunet import (
+    create_local_plans,
+    extract_fingerprint,
+    plan_experiment,
+    preprocess_volume,
+)
 from fl4health_amd.utils.losses import EvaluationLosses, TrainingLosses
+from fl4health_amd.utils.nnunet_utils import (
+    AsyncPatchLoader,
+    NnunetConfig,
+    NnUNetDataLoaderWrapper,
+    PolyLRSchedulerWrapper,
+    convert_deep_supervision_dict_to_list,
+    convert_deep_supervision_list_to_dict,
+    prepare_loss_arg,
+    use_default_signal_handlers,
+)
 
 log = logging.getLogger(__name__)
 
+HAS_NNUNETV2 = importlib.util.find_spec("nnunetv2") is not None
+
 
 class NnunetClient(BasicClient):
-    def __init__(self, *args, max_grad_norm: float = 12.0, **kwargs) -> None:
+    def __init__(
+        self,
+        *args,
+        dataset_name: str = "Dataset999_Local",
+        nnunet_config: str = "3d_fullres",
+        max_grad_norm: float = 12.0,
+        always_preprocess: bool = False,
+        ignore_label: int | None = None,
+        **kwargs,
+    ) -> None:
         super().__init__(*args, **kwargs)
+        self.dataset_name = dataset_name
+        self.nnunet_config = NnunetConfig(nnunet_config)
         self.max_grad_norm = max_grad_norm
+        self.always_preprocess = always_preprocess
+        self.ignore_label = ignore_label
         self.plans: dict | None = None
+        self.plans_name: str | None = None
+        self.fingerprint: dict | None = None
+        self.fingerprint_extracted = False
+        self._preprocessed: tuple[list, list] | None = None
+        self._async_loaders: list = []
 
     # ------------------------------------------------------------------
-    # plans bootstrap (reference nnunet_client.py:388-552 + nnunet_server)
+    # local data hooks (synthetic by default: no network for datasets)
     # ------------------------------------------------------------------
-    def generate_plans(self, config: Config) -> dict:
-        """Derive training plans from the local dataset fingerprint."""
+    def get_local_volumes(self, config: Config) -> tuple[list[torch.Tensor], list[torch.Tensor], list[list[float]]]:
+        """Return (volumes [C, X, Y, Z], segs [X, Y, Z], spacings). Override
+        with a real reader; the default synthesises a small non-trivial
+        dataset shaped by the config."""
+        n = int(config.get("n_train_volumes", 6)) + int(config.get("n_val_volumes", 2))
+        cs = int(config.get("in_channels", 1))
+        ncls = int(config.get("num_classes", 3))
+        lo = int(config.get("min_volume_size", 28))
+        hi = int(config.get("max_volume_size", 44))
+        gen = torch.Generator().manual_seed(abs(hash(self.client_name)) % (2**31))
+        vols, segs, spacings = [], [], []
+        for _ in range(n):
+            shape = [int(torch.randint(lo, hi, (), generator=gen)) for _ in range(3)]
+            v = torch.rand(cs, *shape, generator=gen) + 0.1
+            s = torch.randint(0, ncls, tuple(shape), generator=gen)
+            vols.append(v)
+            segs.append(s)
+            spacings.append([1.0, float(1 + torch.rand((), generator=gen)), 1.0])
+        return vols, segs, spacings
+
+    def get_dataset_json(self, config: Config) -> dict:
+        cs = int(config.get("in_channels", 1))
+        ncls = int(config.get("num_classes", 3))
         return {
-            "patch_size": list(config.get("patch_size", [64, 64, 64])),
-            "in_channels": int(config.get("in_channels", 1)),
-            "num_classes": int(config.get("num_classes", 3)),
-            "base_channels": int(config.get("base_channels", 16)),
-            "num_levels": int(config.get("num_levels", 4)),
+            "name": self.dataset_name,
+            "numTraining": int(config.get("n_train_volumes", 6)),
+            "channel_names": {str(i): f"ch{i}" for i in range(cs)},
+            "labels": {"background": 0, **{f"class{i}": i for i in range(1, ncls)}},
         }
 
-    def get_properties(self, config: Config) -> Config:
-        if config.get("poll_plans", False):
-            self.plans = self.generate_plans(config)
-            return {"nnunet_plans": json.dumps(self.plans)}
-        return super().get_properties(config)
+    # ------------------------------------------------------------------
+    # fingerprint + plans (reference :388-552, :826-885)
+    # ------------------------------------------------------------------
+    def maybe_extract_fingerprint(self, config: Config) -> None:
+        if self.fingerprint_extracted:
+            return
+        vols, segs, spacings = self.get_local_volumes(config)
+        self._local_data = (vols, segs, spacings)
+        self.fingerprint = extract_fingerprint(vols, spacings)
+        self.fingerprint_extracted = True
 
-    def _plans_from_config(self, config: Config) -> dict:
-        if "nnunet_plans" in config:
-            blob = config["nnunet_plans"]
-            if isinstance(blob, bytes):
-                blob = blob.decode()
-            return json.loads(blob)
-        return self.generate_plans(config)
+    @use_default_signal_handlers
+    def _plan_from_local(self, config: Config) -> dict:
+        """Experiment planning for plans election (reference :826-866)."""
+        self.maybe_extract_fingerprint(config)
+        if HAS_NNUNETV2:
+            try:
+                from nnunetv2.experiment_planning.experiment_planners.default_experiment_planner import (  # type: ignore
+                    ExperimentPlanner,
+                )
+
+                planner = ExperimentPlanner(dataset_name_or_id=self.dataset_name, plans_name="temp_plans")
+                plans = planner.plan_experiment()
+                plans["plans_name"] = self.dataset_name + "_plans"
+                return plans
+            except Exception:  # noqa: BLE001 — fall back to the native planner
+                log.exception("nnunetv2 planner failed; using the native planner")
+        assert self.fingerprint is not None
+        plans = plan_experiment(
+            self.fingerprint,
+            self.get_dataset_json(config),
+            max_patch_voxels=int(config.get("max_patch_voxels", 64 ** 3)),
+            base_channels=int(config.get("base_channels", 32)),
+            max_levels=int(config.get("num_levels", 5)),
+        )
+        plans["plans_name"] = self.dataset_name + "_plans"
+        return plans
+
+    def create_plans(self, config: Config) -> dict:
+        """Localise the server-elected plans (reference :388-495)."""
+        blob = config["nnunet_plans"]
+        source = pickle.loads(blob) if isinstance(blob, bytes) else blob
+        assert self.fingerprint is not None
+        plans = create_local_plans(
+            source, self.fingerprint, self.get_dataset_json(config), self.dataset_name,
+            plans_name=self.plans_name,
+        )
+        self.plans_name = plans["plans_name"]
+        return plans
+
+    @use_default_signal_handlers
+    def get_properties(self, config: Config) -> Config:
+        if "nnunet_plans" not in config:
+            log.info("initializing global nnunet plans from the local dataset")
+            config = dict(config)
+            config["nnunet_plans"] = pickle.dumps(self._plan_from_local(config))
+        props = dict(super().get_properties(config))
+        props["nnunet_plans"] = config["nnunet_plans"]
+        if not self.initialized:
+            self.setup_client(config)
+        assert self.plans is not None
+        net = self._network_params()
+        props["num_input_channels"] = net["in_channels"]
+        props["num_segmentation_heads"] = net["num_classes"]
+        props["enable_deep_supervision"] = True
+        return props
 
     # ------------------------------------------------------------------
+    @use_default_signal_handlers
+    def setup_client(self, config: Config) -> None:
+        if self.device.type == "cuda":
+            torch.cuda.empty_cache()  # reference empty_cache :800-806
+        self.maybe_extract_fingerprint(config)
+        if self.plans is None:
+            if "nnunet_plans" in config:
+                self.plans = self.create_plans(config)
+            else:
+                self.plans = self._plan_from_local(config)
+        self.maybe_preprocess(config)
+        super().setup_client(config)
+
+    def maybe_preprocess(self, config: Config) -> None:
+        """Preprocess local volumes per the (localised) plans (reference
+        maybe_preprocess :488-520)."""
+        if self._preprocessed is not None and not self.always_preprocess:
+            return
+        assert self.plans is not None
+        cfg = self.plans["configurations"][self.nnunet_config.value]
+        props = self.plans["foreground_intensity_properties_per_channel"]
+        vols, segs, spacings = self._local_data
+        pv, ps = [], []
+        for v, s, sp in zip(vols, segs, spacings):
+            out_v, out_s = preprocess_volume(v, s, sp, cfg["spacing"], props)
+            pv.append(out_v)
+            ps.append(out_s)
+        self._preprocessed = (pv, ps)
+
+    def _network_params(self) -> dict:
+        assert self.plans is not None
+        if "network" in self.plans:
+            return self.plans["network"]
+        cfg = self.plans["configurations"][self.nnunet_config.value]
+        djson = self.get_dataset_json({})
+        return {
+            "in_channels": len(djson["channel_names"]),
+            "num_classes": len(djson["labels"]),
+            "base_channels": int(cfg.get("UNet_base_num_features", 32)),
+            "num_levels": int(cfg.get("n_stages", 4)),
+        }
+
     def get_model(self, config: Config) -> torch.nn.Module:
-        self.plans = self._plans_from_config(config)
+        net = self._network_params()
         return UNet3D(
-            in_channels=self.plans["in_channels"],
-            num_classes=self.plans["num_classes"],
-            base_channels=self.plans["base_channels"],
-            num_levels=self.plans["num_levels"],
+            in_channels=net["in_channels"],
+            num_classes=net["num_classes"],
+            base_channels=min(net["base_channels"], 32),
+            num_levels=net["num_levels"],
             deep_supervision=True,
         )
 
     def get_data_loaders(self, config: Config) -> tuple[DataLoader, DataLoader | None]:
-        """Synthetic volumes shaped by the plans (no network for datasets)."""
-        assert self.plans is not None
-        ps = self.plans["patch_size"]
-        n_train = int(config.get("n_train_volumes", 8))
-        n_val = int(config.get("n_val_volumes", 2))
-        bs = int(config.get("batch_size", 2))
-        gen = torch.Generator().manual_seed(hash(self.client_name) % (2**31))
-        x = torch.randn(n_train + n_val, self.plans["in_channels"], *ps, generator=gen)
-        y = torch.randint(0, self.plans["num_classes"], (n_train + n_val, *ps), generator=gen)
-        train = TensorDataset(x[:n_train], y[:n_train])
-        val = TensorDataset(x[n_train:], y[n_train:])
-        return DataLoader(train, batch_size=bs, shuffle=True), DataLoader(val, batch_size=bs)
+        assert self.plans is not None and self._preprocessed is not None
+        cfg = self.plans["configurations"][self.nnunet_config.value]
+        n_train = int(config.get("n_train_volumes", 6))
+        pv, ps = self._preprocessed
+        net = self._network_params()
+        train_aug = AsyncPatchLoader(
+            pv[:n_train], ps[:n_train], cfg["patch_size"], int(cfg["batch_size"]),
+            n_batches_per_epoch=int(config.get("n_batches_per_epoch", 10)),
+            deep_supervision_levels=net["num_levels"],
+            seed=abs(hash(self.client_name)) % (2**31),
+        )
+        val_aug = AsyncPatchLoader(
+            pv[n_train:] or pv[:1], ps[n_train:] or ps[:1], cfg["patch_size"], int(cfg["batch_size"]),
+            n_batches_per_epoch=int(config.get("n_val_batches", 2)),
+            deep_supervision_levels=1,
+            seed=1 + abs(hash(self.client_name)) % (2**31),
+        )
+        train = NnUNetDataLoaderWrapper(train_aug, self.nnunet_config)
+        val = NnUNetDataLoaderWrapper(val_aug, self.nnunet_config)
+        self._async_loaders = [train, val]
+        return train, val
 
     def get_optimizer(self, config: Config):
-        return torch.optim.SGD(self.model.parameters(), lr=float(config.get("lr", 1e-2)), momentum=0.99, nesterov=True, weight_decay=3e-5)
+        return torch.optim.SGD(
+            self.model.parameters(), lr=float(config.get("lr", 1e-2)),
+            momentum=0.99, nesterov=True, weight_decay=3e-5,
+        )
 
     def get_lr_scheduler(self, optimizer_key: str, config: Config):
-        max_steps = int(config.get("n_server_rounds", 10)) * int(config.get("local_steps", 10))
-        return PolyLRScheduler(self.optimizers[optimizer_key], float(config.get("lr", 1e-2)), max_steps)
+        max_steps = max(1, int(config.get("n_server_rounds", 10)) * int(config.get("local_steps", 10)))
+        return PolyLRSchedulerWrapper(self.optimizers[optimizer_key], float(config.get("lr", 1e-2)), max_steps)
 
     def get_criterion(self, config: Config) -> torch.nn.Module:
-        assert self.plans is not None
-        return DeepSupervisionLoss(self.plans["num_classes"])
+        return DeepSupervisionLoss(self._network_params()["num_classes"])
 
     # ------------------------------------------------------------------
-    def predict(self, input):
-        output = self.model(input)
-        if isinstance(output, list):
-            # deep supervision: highest-resolution head is "the" prediction,
-            # full pyramid kept for the loss
-            return {"prediction": output[0]}, {"ds_outputs": output}
-        return {"prediction": output}, {}
+    # deep supervision plumbing (reference :625-798)
+    # ------------------------------------------------------------------
+    def predict(self, input) -> tuple[TorchPredType, dict[str, torch.Tensor]]:
+        if not isinstance(input, torch.Tensor):
+            raise TypeError('"input" must be a torch.Tensor for NnunetClient')
+        if self.device.type == "cuda":
+            with torch.autocast(self.device.type, enabled=True):
+                output = self.model(input)
+        else:
+            output = self.model(input)
+        if isinstance(output, torch.Tensor):
+            return {"prediction": output}, {}
+        if isinstance(output, (list, tuple)):
+            from fl4health_amd.utils.nnunet_utils import NNUNET_N_SPATIAL_DIMS
+
+            nd = NNUNET_N_SPATIAL_DIMS[self.nnunet_config]
+            return convert_deep_supervision_list_to_dict(list(output), nd), {}
+        raise TypeError("unexpected nnunet model output type")
 
     def compute_loss_and_additional_losses(self, preds: TorchPredType, features, target: TorchTargetType):
-        outputs = features.get("ds_outputs", preds["prediction"])
-        return self.criterion(outputs, target), None
+        loss_preds = prepare_loss_arg(preds)
+        loss_targets = prepare_loss_arg(target) if not isinstance(target, torch.Tensor) else target
+        if isinstance(loss_preds, list) and isinstance(loss_targets, list):
+            assert len(loss_preds) == len(loss_targets), (
+                f"{len(loss_preds)} predictions vs {len(loss_targets)} targets"
+            )
+            # clip the target pyramid to the heads the model actually emits
+            loss_targets = loss_targets[: len(loss_preds)]
+        if self.device.type == "cuda":
+            with torch.autocast(self.device.type, enabled=True):
+                return self.criterion(loss_preds, loss_targets), None
+        return self.criterion(loss_preds, loss_targets), None
+
+    def mask_data(self, pred: torch.Tensor, target: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+        """Ignore-label masking (reference :705-758)."""
+        assert self.ignore_label is not None
+        mask = (target != self.ignore_label).float()
+        new_target = target.clone()
+        new_target[new_target == self.ignore_label] = 0
+        if mask.dim() == pred.dim() - 1:
+            mask = mask.unsqueeze(1)
+        mask = mask.expand_as(pred) if mask.shape != pred.shape else mask
+        return pred * mask, new_target
+
+    def update_metric_manager(self, preds: TorchPredType, target: TorchTargetType, metric_manager: MetricManager) -> None:
+        if len(preds) > 1:
+            m_pred = convert_deep_supervision_dict_to_list(preds)[0]
+        else:
+            m_pred = next(iter(preds.values()))
+        if isinstance(target, dict):
+            m_target = (
+                convert_deep_supervision_dict_to_list(target)[0] if len(target) > 1 else next(iter(target.values()))
+            )
+        else:
+            m_target = target
+        if self.ignore_label is not None:
+            m_pred, m_target = self.mask_data(m_pred, m_target)
+        metric_manager.update({"prediction": m_pred}, m_target)
 
     def transform_gradients(self, losses: TrainingLosses) -> None:
         torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.max_grad_norm)
 
     def compute_evaluation_loss(self, preds: TorchPredType, features, target: TorchTargetType) -> EvaluationLosses:
         with torch.no_grad():
-            loss = self.criterion(preds["prediction"], target)
+            loss, _ = self.compute_loss_and_additional_losses(preds, features, target)
         return EvaluationLosses(checkpoint=loss)
+
+    # ------------------------------------------------------------------
+    # lifecycle (reference :883-935)
+    # ------------------------------------------------------------------
+    def update_before_train(self, current_server_round: int) -> None:
+        gc.collect()
+        if current_server_round == 2:
+            gc.freeze()  # reference: collect runs faster frozen after round 1
+
+    def shutdown(self) -> None:
+        gc.unfreeze()
+        gc.collect()
+        for dl in self._async_loaders:
+            try:
+                dl.shutdown()
+            except Exception:  # noqa: BLE001
+                log.exception("loader shutdown failed")
+        self._async_loaders = []
+        super().shutdown()

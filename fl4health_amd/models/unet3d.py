@@ -98,17 +98,25 @@ class DeepSupervisionLoss(nn.Module):
         ce = Fn.cross_entropy(logits, target)
         return self.ce_weight * ce + self.dice_weight * self._dice_loss(logits, target)
 
-    def forward(self, outputs, target: torch.Tensor) -> torch.Tensor:
+    def forward(self, outputs, target) -> torch.Tensor:
+        """``target`` may be a single full-resolution map (downsampled here
+        per scale) or an already-built pyramid list matching ``outputs``
+        (the nnU-Net loader protocol, reference nnunet_client.py:659-706)."""
         if isinstance(outputs, torch.Tensor):
-            return self._single(outputs, target)
+            t = target[0] if isinstance(target, (list, tuple)) else target
+            return self._single(outputs, t)
+        targets = list(target) if isinstance(target, (list, tuple)) else None
         weights = [0.5**i for i in range(len(outputs))]
         wsum = sum(weights)
         total = torch.zeros((), device=outputs[0].device)
-        for w, logits in zip(weights, outputs):
-            scale_target = target
-            if logits.shape[2:] != target.shape[1:]:
+        for i, (w, logits) in enumerate(zip(weights, outputs)):
+            if targets is not None and i < len(targets):
+                scale_target = targets[i]
+            else:
+                scale_target = targets[0] if targets is not None else target
+            if logits.shape[2:] != scale_target.shape[1:]:
                 scale_target = (
-                    Fn.interpolate(target.unsqueeze(1).float(), size=logits.shape[2:], mode="nearest")
+                    Fn.interpolate(scale_target.unsqueeze(1).float(), size=logits.shape[2:], mode="nearest")
                     .squeeze(1)
                     .long()
                 )
