@@ -88,7 +88,9 @@ def test_feature_alignment_cross_schema():
     from fl4health_amd.feature_alignment.tab_features_info_encoder import TabularFeaturesInfoEncoder
     from fl4health_amd.feature_alignment.tab_features_preprocessor import TabularFeaturesPreprocessor
 
-    df_a = pd.DataFrame({"num": [1.0, 2.0, 3.0], "cat": ["x", "y", "x"], "label": [0, 1, 0]})
+    df_a = pd.DataFrame(
+        {"num": [1.0, 2.0, 3.0], "cat": ["x", "y", "z"], "label": [0, 1, 0]}
+    )
     df_b = pd.DataFrame({"num": [5.0, 6.0], "label": [1, 0]})  # missing 'cat'
     enc = TabularFeaturesInfoEncoder.encoder_from_dataframe(df_a, None, "label")
     enc2 = TabularFeaturesInfoEncoder.from_json(enc.to_json())
@@ -96,7 +98,9 @@ def test_feature_alignment_cross_schema():
     xa, ya = pre.preprocess(df_a)
     xb, yb = pre.preprocess(df_b)
     assert xa.shape[1] == xb.shape[1] == enc.input_dimension()
-    assert (xb[:, -2:] == 0).all()  # missing categorical -> zero block
+    # 'cat' is ORDINAL (3 categories) -> one-hot block of 3; the fill value
+    # "UNKNOWN" is outside the vocabulary -> all-zero block (unknown ignored)
+    assert (xb[:, :3] == 0).all()
 
 
 def test_metrics_utils_alignment():
