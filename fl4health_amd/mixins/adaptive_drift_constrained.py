@@ -34,6 +34,9 @@ class BaseFlexibleMixin:
 
     def __init_subclass__(cls, **kwargs) -> None:
         super().__init_subclass__(**kwargs)
+        # pure mixin classes (not yet composed with a client) are exempt
+        if cls.__name__.endswith("Mixin") or cls.__name__.startswith("_"):
+            return
         if not any(issubclass(b, BasicClient) for b in cls.__mro__ if b not in (cls, BaseFlexibleMixin, object)):
             import warnings
 
@@ -57,9 +60,8 @@ def apply_adaptive_drift_to_client(client_cls: type[BasicClient]) -> type[BasicC
 
 
 def make_it_personal(client_cls: type[BasicClient], mode: str = "ditto") -> type[BasicClient]:
-    """Personalization factory (reference personalized/ditto.py / mr_mtl.py)."""
-    from fl4health_amd.clients.adaptive_drift_constraint_client import MrMtlClient
-    from fl4health_amd.clients.ditto_client import DittoClient
+    """Personalization factory (reference personalized/__init__.py:19-41);
+    the hook-driven implementation lives in mixins/personalized.py."""
+    from fl4health_amd.mixins.personalized import make_it_personal as _factory
 
-    base = DittoClient if mode == "ditto" else MrMtlClient
-    return type(f"{mode.title()}{client_cls.__name__}", (base, client_cls), {})
+    return _factory(client_cls, mode)
