@@ -33,7 +33,10 @@ class FedAvgWithAdaptiveConstraint(BasicFedAvg):
         weighted_train_losses: bool = False,
         **kwargs,
     ) -> None:
-        assert kwargs.get("initial_parameters") is not None, "initial parameters are required for this strategy"
+        # initial parameters may be None: the server then polls one client for
+        # initial weights and add_auxiliary_information appends mu afterwards
+        # (reference base_server.py:516-541; needed for late-architecture
+        # workloads like nnU-Net where plans election precedes model shape)
         super().__init__(**kwargs)
         self.loss_weight = initial_loss_weight
         self.adapt_loss_weight = adapt_loss_weight

@@ -264,3 +264,23 @@ def test_mkmmd_client_trains_on_gpu():
     assert len(hist.losses_distributed) == 2
     for _, loss in hist.losses_distributed:
         assert torch.isfinite(torch.tensor(loss))
+
+
+@requires_gpu
+def test_gpu_heavy_workload_examples():
+    """BASELINE heavy configs on hardware: one round each of the BERT + LoRA
+    + MOON config and the 3D U-Net FedBN config through the example
+    entry points (small shapes; the --full shapes are bench-only)."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    for mod in ("heavy_workloads.bert_moon_lora", "heavy_workloads.unet3d_fedbn"):
+        out = subprocess.run(
+            [sys.executable, "-m", f"examples.{mod}", "--rounds", "1", "--local_steps", "2", "--batch_size", "4"],
+            capture_output=True, text=True, timeout=600, cwd=str(root),
+            env={**__import__("os").environ, "PYTHONPATH": str(root)},
+        )
+        assert out.returncode == 0, (mod, out.stderr[-2000:])
+        assert "[SUMMARY]" in out.stdout, mod
