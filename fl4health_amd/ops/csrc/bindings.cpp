@@ -29,6 +29,8 @@ void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, con
 void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int, int, int, int,
                              int, hipStream_t);
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
+void launch_moon_contrastive(const float*, const float*, const float*, int, int, int64_t, float,
+                             float*, float*, hipStream_t);
 void launch_conv3x3_fwd(const void*, const void*, const float*, void*, int, int, int, int, int,
                         hipStream_t);
 void launch_mkmmd_sums(const float*, const float*, double*, float*, int, int64_t, int64_t, int,
@@ -325,6 +327,28 @@ torch::Tensor pack_kb32(torch::Tensor w, bool bwd) {
   return out;
 }
 
+// Fused MOON contrastive loss + input gradient (K8): one workgroup per
+// sample computes the cosine-similarity logits against [pos | negs], the
+// softmax-CE loss, and dz in a single launch (partners are frozen snapshots).
+std::vector<torch::Tensor> moon_contrastive(torch::Tensor z, torch::Tensor pos,
+                                            torch::Tensor neg, double tau) {
+  check_f32(z, "z");
+  check_f32(pos, "pos");
+  check_f32(neg, "neg");
+  TORCH_CHECK(z.dim() == 2 && pos.sizes() == z.sizes(), "z/pos must be [B, D]");
+  TORCH_CHECK(neg.dim() == 3 && neg.size(1) == z.size(0) && neg.size(2) == z.size(1),
+              "neg must be [K, B, D]");
+  int K = (int)neg.size(0);
+  TORCH_CHECK(K + 1 <= 17, "at most 16 negative pairs");
+  int B = (int)z.size(0);
+  auto loss = torch::empty({B}, z.options());
+  auto dz = torch::empty_like(z);
+  launch_moon_contrastive(z.data_ptr<float>(), pos.data_ptr<float>(), neg.data_ptr<float>(), K, B,
+                          z.size(1), (float)(1.0 / tau), loss.data_ptr<float>(),
+                          dz.data_ptr<float>(), stream());
+  return {loss, dz};
+}
+
 // Fused multi-bandwidth Gaussian-kernel sums over a pairwise-distance Gram
 // (reference losses/mkmmd_loss.py:96-135).
 torch::Tensor mkmmd_sums(torch::Tensor d, torch::Tensor gammas, bool skip_diag) {
@@ -377,6 +401,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_fwd_kb32", &conv3x3_fwd_kb32,
         "direct 3x3 NHWC bf16 conv forward, KB=32 glds-pipelined variant");
   m.def("pack_kb32", &pack_kb32, "fused weight pack for conv3x3_fwd_kb32");
+  m.def("moon_contrastive", &moon_contrastive,
+        "fused MOON contrastive loss + dz (cosine logits, softmax-CE, label 0)");
   m.def("mkmmd_sums", &mkmmd_sums, "per-bandwidth Gaussian kernel sums over a Gram");
   m.def("mkmmd_backward", &mkmmd_backward, "dL/dGram for mkmmd_sums");
 }

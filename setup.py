@@ -25,6 +25,7 @@ setup(
                 os.path.join(CSRC, "bn_ops.hip"),
                 os.path.join(CSRC, "mmd_ops.hip"),
                 os.path.join(CSRC, "conv_ops.hip"),
+                os.path.join(CSRC, "contrastive_ops.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3"],

@@ -448,3 +448,33 @@ def test_cdna_conv2d_force_mfma_all_widths():
         y_ours.backward(gy.to(torch.bfloat16))
         r = (x2.grad.float() - x1.grad.float()).abs().max() / x1.grad.float().abs().max().clamp(min=1e-6)
         assert r < 5e-2, float(r)
+
+
+@requires_gpu
+def test_moon_contrastive_fused_matches_eager():
+    """Fused K8 kernel (cosine logits + softmax-CE + dz in one launch) vs the
+    eager MoonContrastiveLoss oracle, value AND gradient."""
+    from fl4health_amd.losses.contrastive_loss import MoonContrastiveLoss, _FusedMoonContrastiveFn
+
+    torch.manual_seed(0)
+    loss_mod = MoonContrastiveLoss(temperature=0.5)
+    for b, d, k in [(8, 64, 1), (32, 768, 3), (5, 33, 2)]:
+        z1 = torch.randn(b, d, device="cuda", requires_grad=True)
+        z2 = z1.detach().clone().requires_grad_(True)
+        pos = torch.randn(1, b, d, device="cuda")
+        neg = torch.randn(k, b, d, device="cuda")
+        # eager oracle (force the torch path by making pos require grad... use internals)
+        logits = torch.nn.functional.cosine_similarity(z1, pos[0], dim=-1).reshape(-1, 1)
+        negs = torch.nn.functional.cosine_similarity(z1.unsqueeze(0).expand(k, -1, -1), neg, dim=-1)
+        eager = torch.nn.functional.cross_entropy(
+            torch.cat((logits, negs.T), dim=1) / 0.5,
+            torch.zeros(b, dtype=torch.long, device="cuda"),
+        )
+        eager.backward()
+        fused = _FusedMoonContrastiveFn.apply(z2, pos[0], neg, 0.5)
+        fused.backward()
+        assert torch.allclose(fused, eager, atol=1e-5), (b, d, k, float(fused), float(eager))
+        assert torch.allclose(z2.grad, z1.grad, atol=1e-5), (b, d, k, float((z2.grad - z1.grad).abs().max()))
+        # the module-level adoption path picks the kernel on GPU
+        out = loss_mod(z2.detach().requires_grad_(True), pos, neg)
+        assert torch.allclose(out, eager, atol=1e-5)
