@@ -28,7 +28,12 @@ class Client(FedBnClient):
         self.base_ch = 32 if args.full else 4
 
     def get_model(self, config):
-        return UNet3D(1, 3, base_channels=self.base_ch, num_levels=self.levels, deep_supervision=True)
+        model = UNet3D(1, 3, base_channels=self.base_ch, num_levels=self.levels, deep_supervision=True)
+        if torch.cuda.is_available():
+            from fl4health_amd.ops.instancenorm import fuse_unet3d_norm_relu
+
+            model = fuse_unet3d_norm_relu(model)
+        return model
 
     def get_data_loaders(self, config):
         gen = torch.Generator().manual_seed(self.seed)

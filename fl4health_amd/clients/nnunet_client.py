@@ -230,13 +230,22 @@ class NnunetClient(BasicClient):
 
     def get_model(self, config: Config) -> torch.nn.Module:
         net = self._network_params()
-        return UNet3D(
+        model = UNet3D(
             in_channels=net["in_channels"],
             num_classes=net["num_classes"],
             base_channels=min(net["base_channels"], 32),
             num_levels=net["num_levels"],
             deep_supervision=True,
         )
+        if self.device.type == "cuda":
+            # MIOpen solver find must be on BEFORE the first conv (cached per
+            # process; 3D U-Net default picks are the im2col fallback) and the
+            # norm+activation runs the fused CDNA kernels
+            torch.backends.cudnn.benchmark = True
+            from fl4health_amd.ops.instancenorm import fuse_unet3d_norm_relu
+
+            model = fuse_unet3d_norm_relu(model)
+        return model
 
     def get_data_loaders(self, config: Config) -> tuple[DataLoader, DataLoader | None]:
         assert self.plans is not None and self._preprocessed is not None

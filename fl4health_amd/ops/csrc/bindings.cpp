@@ -31,6 +31,11 @@ void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int,
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
 void launch_moon_contrastive(const float*, const float*, const float*, int, int, int64_t, float,
                              float*, float*, hipStream_t);
+void launch_in3d_fwd(const void*, void*, float*, float*, float*, const float*, const float*, int,
+                     int, int64_t, int, float, float, hipStream_t);
+void launch_in3d_bwd(const void*, const void*, void*, float*, float*, float*, float*, float*,
+                     const float*, const float*, const float*, const float*, int, int, int64_t,
+                     int, float, hipStream_t);
 void launch_conv3x3_fwd(const void*, const void*, const float*, void*, int, int, int, int, int,
                         hipStream_t);
 void launch_mkmmd_sums(const float*, const float*, double*, float*, int, int64_t, int64_t, int,
@@ -349,6 +354,61 @@ std::vector<torch::Tensor> moon_contrastive(torch::Tensor z, torch::Tensor pos,
   return {loss, dz};
 }
 
+static int in3d_slices(int64_t planes, int64_t L) {
+  // enough (plane, slice) blocks to fill 256 CUs; cap the partial buffer
+  int S = (int)std::min<int64_t>(64, std::max<int64_t>(1, (1024 + planes - 1) / planes));
+  while ((int64_t)S > 1 && (L + S - 1) / S < 4096) --S;  // keep chunks >= 4096 elems
+  return std::max(S, 1);
+}
+
+// Fused InstanceNorm3d + LeakyReLU forward over NCDHW bf16 planes.
+// Returns (y, mean[P], invstd[P]) — stats saved for backward.
+std::vector<torch::Tensor> in3d_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
+                                    double eps, double slope) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 5 &&
+              x.scalar_type() == torch::kBFloat16, "x must be [N,C,D,H,W] bf16 contiguous");
+  check_f32(gamma, "gamma");
+  check_f32(beta, "beta");
+  int64_t N = x.size(0), C = x.size(1);
+  int64_t L = x.size(2) * x.size(3) * x.size(4);
+  int64_t P = N * C;
+  int S = in3d_slices(P, L);
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
+  auto mean = torch::empty({P}, opts);
+  auto invstd = torch::empty({P}, opts);
+  auto partial = torch::empty({P * S * 2}, opts);
+  auto y = torch::empty_like(x);
+  launch_in3d_fwd(x.data_ptr(), y.data_ptr(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                  partial.data_ptr<float>(), gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                  (int)P, (int)C, L, S, (float)eps, (float)slope, stream());
+  return {y, mean, invstd};
+}
+
+// Backward: returns (dx, dgamma, dbeta).
+std::vector<torch::Tensor> in3d_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
+                                    torch::Tensor invstd, torch::Tensor gamma,
+                                    torch::Tensor beta, double slope) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16 && dy.scalar_type() == torch::kBFloat16);
+  int64_t N = x.size(0), C = x.size(1);
+  int64_t L = x.size(2) * x.size(3) * x.size(4);
+  int64_t P = N * C;
+  int S = in3d_slices(P, L);
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
+  auto partial = torch::empty({P * S * 2}, opts);
+  auto s1 = torch::empty({P}, opts);
+  auto s2 = torch::empty({P}, opts);
+  auto dgamma = torch::empty({C}, opts);
+  auto dbeta = torch::empty({C}, opts);
+  auto dx = torch::empty_like(x);
+  launch_in3d_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), partial.data_ptr<float>(),
+                  s1.data_ptr<float>(), s2.data_ptr<float>(), dgamma.data_ptr<float>(),
+                  dbeta.data_ptr<float>(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(), (int)P, (int)C, L, S,
+                  (float)slope, stream());
+  return {dx, dgamma, dbeta};
+}
+
 // Fused multi-bandwidth Gaussian-kernel sums over a pairwise-distance Gram
 // (reference losses/mkmmd_loss.py:96-135).
 torch::Tensor mkmmd_sums(torch::Tensor d, torch::Tensor gammas, bool skip_diag) {
@@ -403,6 +463,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_kb32", &pack_kb32, "fused weight pack for conv3x3_fwd_kb32");
   m.def("moon_contrastive", &moon_contrastive,
         "fused MOON contrastive loss + dz (cosine logits, softmax-CE, label 0)");
+  m.def("in3d_fwd", &in3d_fwd, "fused InstanceNorm3d + LeakyReLU forward (NCDHW bf16)");
+  m.def("in3d_bwd", &in3d_bwd, "fused InstanceNorm3d + LeakyReLU backward");
   m.def("mkmmd_sums", &mkmmd_sums, "per-bandwidth Gaussian kernel sums over a Gram");
   m.def("mkmmd_backward", &mkmmd_backward, "dL/dGram for mkmmd_sums");
 }

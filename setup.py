@@ -26,6 +26,7 @@ setup(
                 os.path.join(CSRC, "mmd_ops.hip"),
                 os.path.join(CSRC, "conv_ops.hip"),
                 os.path.join(CSRC, "contrastive_ops.hip"),
+                os.path.join(CSRC, "in_ops.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3"],

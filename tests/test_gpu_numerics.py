@@ -478,3 +478,52 @@ def test_moon_contrastive_fused_matches_eager():
         # the module-level adoption path picks the kernel on GPU
         out = loss_mod(z2.detach().requires_grad_(True), pos, neg)
         assert torch.allclose(out, eager, atol=1e-5)
+
+
+@requires_gpu
+def test_fused_instance_norm3d_leaky_relu():
+    """Fused IN3d+LeakyReLU kernels vs the fp32 torch oracle: value, dx,
+    dgamma, dbeta (deterministic two-stage reductions)."""
+    import torch.nn.functional as Fn
+
+    from fl4health_amd.ops.instancenorm import _FusedIN3dFn
+
+    torch.manual_seed(0)
+    for n, c, s in [(2, 8, 12), (1, 32, 20), (3, 4, 9)]:
+        x16 = (torch.randn(n, c, s, s, s, device="cuda") * 2).to(torch.bfloat16)
+        x1 = x16.float().requires_grad_(True)
+        g1 = torch.randn(c, device="cuda", requires_grad=True)
+        b1 = torch.randn(c, device="cuda", requires_grad=True)
+        ref = Fn.leaky_relu(Fn.instance_norm(x1, weight=g1, bias=b1, eps=1e-5), 0.01)
+        x2 = x16.clone().requires_grad_(True)
+        g2 = g1.detach().clone().requires_grad_(True)
+        b2 = b1.detach().clone().requires_grad_(True)
+        out = _FusedIN3dFn.apply(x2, g2, b2, 1e-5, 0.01)
+        rel = (out.float() - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
+        assert rel < 3e-2, float(rel)
+        gy = torch.randn_like(ref)
+        ref.backward(gy)
+        out.backward(gy.to(torch.bfloat16))
+        for name, (a, bb) in {
+            "dx": (x1.grad, x2.grad.float()),
+            "dgamma": (g1.grad, g2.grad),
+            "dbeta": (b1.grad, b2.grad),
+        }.items():
+            r = (a - bb).abs().max() / a.abs().max().clamp(min=1e-5)
+            assert r < 5e-2, (name, float(r), n, c, s)
+
+
+@requires_gpu
+def test_fused_convblock3d_trains():
+    from fl4health_amd.models.unet3d import UNet3D
+    from fl4health_amd.ops.instancenorm import FusedConvBlock3d, fuse_unet3d_norm_relu
+
+    torch.manual_seed(0)
+    model = fuse_unet3d_norm_relu(UNet3D(1, 2, base_channels=4, num_levels=2)).cuda()
+    assert any(type(m) is FusedConvBlock3d for m in model.modules())
+    x = torch.randn(2, 1, 16, 16, 16, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = model(x)
+        loss = sum(o.float().pow(2).mean() for o in out) if isinstance(out, list) else out.float().pow(2).mean()
+    loss.backward()
+    assert all(p.grad is not None for p in model.parameters() if p.requires_grad)

@@ -62,11 +62,15 @@ def bench_bert():
     log(f"BERT-base + LoRA + MOON (batch {batch}, seq {seq}, bf16): {ms:.1f} ms/step = {toks/1e3:.1f}k tokens/s/GPU")
 
 
-def bench_unet(patch=128, base=32, levels=5, batch=2, channels_last=False):
+def bench_unet(patch=128, base=32, levels=5, batch=2, channels_last=False, fused_in=True):
     from fl4health_amd.models.unet3d import DeepSupervisionLoss, UNet3D
 
     torch.manual_seed(0)
     model = UNet3D(1, 3, base_channels=base, num_levels=levels, deep_supervision=True).cuda()
+    if fused_in:
+        from fl4health_amd.ops.instancenorm import fuse_unet3d_norm_relu
+
+        model = fuse_unet3d_norm_relu(model)
     if channels_last:
         model = model.to(memory_format=torch.channels_last_3d)
     model.train()
@@ -96,7 +100,7 @@ def bench_unet(patch=128, base=32, levels=5, batch=2, channels_last=False):
     torch.cuda.synchronize()
     ms = (time.perf_counter() - t0) / n * 1e3
     vox = batch * patch**3 / (ms / 1e3)
-    log(f"3D U-Net {patch}^3 (base {base}, {levels} levels, batch {batch}, bf16, ds, cl={channels_last}): "
+    log(f"3D U-Net {patch}^3 (base {base}, {levels} levels, batch {batch}, bf16, ds, cl={channels_last}, fusedIN={fused_in}): "
         f"{ms:.1f} ms/step = {vox/1e6:.1f}M voxels/s/GPU; peak mem {torch.cuda.max_memory_allocated()/2**30:.1f} GiB")
 
 
