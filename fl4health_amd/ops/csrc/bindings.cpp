@@ -187,6 +187,19 @@ void clip_rowsum_(torch::Tensor g, torch::Tensor sqnorms, torch::Tensor out, dou
                      (float)clip_bound, B, D, stream());
 }
 
+// fused clipped-rowsum + DP Gaussian noise (one pass; Philox stream matches
+// gaussian_noise_ for the same seed/offset)
+void clip_rowsum_noise_(torch::Tensor g, torch::Tensor sqnorms, torch::Tensor out,
+                        double clip_bound, double sigma, int64_t seed, int64_t offset) {
+  check_f32(g, "g"); check_f32(sqnorms, "sqnorms"); check_f32(out, "out");
+  int64_t B = g.size(0);
+  int64_t D = g.numel() / B;
+  TORCH_CHECK(out.numel() == D, "out must be [D]");
+  launch_clip_rowsum_noise(g.data_ptr<float>(), sqnorms.data_ptr<float>(), out.data_ptr<float>(),
+                           (float)clip_bound, (float)sigma, (uint64_t)seed, (uint64_t)offset, B, D,
+                           stream());
+}
+
 void confusion_counts_(torch::Tensor preds, torch::Tensor targets, torch::Tensor out) {
   TORCH_CHECK(preds.is_cuda() && preds.scalar_type() == torch::kInt64 && preds.is_contiguous());
   TORCH_CHECK(targets.is_cuda() && targets.scalar_type() == torch::kInt64 && targets.is_contiguous());
@@ -454,6 +467,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gaussian_noise_", &gaussian_noise_, "x = a*x + sigma*N(0,1), philox");
   m.def("bernoulli_mask", &bernoulli_mask, "bernoulli(sigmoid(scores)) mask (+masked weight)");
   m.def("per_sample_sqnorm_", &per_sample_sqnorm_, "accumulate per-sample grad sq norms");
+  m.def("clip_rowsum_noise_", &clip_rowsum_noise_,
+        "fused clipped rowsum + DP Gaussian noise (single pass over the grads)");
   m.def("clip_rowsum_", &clip_rowsum_, "clipped per-sample grad sum");
   m.def("confusion_counts_", &confusion_counts_, "streaming TP/FP/FN/TN counts");
   m.def("weighted_sum_rows", &weighted_sum_rows, "out = sum_k w[k]*stack[k]");

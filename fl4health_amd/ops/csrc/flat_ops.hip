@@ -433,6 +433,44 @@ extern "C" void launch_clip_rowsum(const float* g, const float* sqnorms, float* 
   clip_rowsum_kernel<<<grid_1d(D), BLOCK, 0, s>>>(g, sqnorms, out, clip_bound, B, D);
 }
 
+// Fused clip_rowsum + Gaussian noise (K7 epilogue): the clipped per-sample
+// sum and the DP noise land in one pass instead of re-reading the whole grad
+// in a second kernel. Philox addressing matches gaussian_noise_kernel
+// (4 elements per counter block) so the noise stream is identical to the
+// unfused path for the same (seed, offset).
+__global__ __launch_bounds__(BLOCK) void clip_rowsum_noise_kernel(
+    const float* __restrict__ g, const float* __restrict__ sqnorms, float* __restrict__ out,
+    float clip_bound, float sigma, uint64_t seed, uint64_t offset, int64_t B, int64_t D) {
+  int64_t nblk = (D + 3) / 4;
+  GSL(q, nblk, STRIDE) {
+    Philox4 r = philox4x32(seed, offset + (uint64_t)q);
+    float z0, z1, z2, z3;
+    box_muller(r.x, r.y, &z0, &z1);
+    box_muller(r.z, r.w, &z2, &z3);
+    float zs[4] = {z0, z1, z2, z3};
+    int64_t base = q * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int64_t d = base + j;
+      if (d >= D) continue;
+      float acc = 0.0f;
+      for (int64_t b = 0; b < B; ++b) {
+        float nrm = sqrtf(sqnorms[b]) + 1e-6f;
+        float coef = (nrm > clip_bound) ? clip_bound / nrm : 1.0f;
+        acc = fmaf(coef, g[b * D + d], acc);
+      }
+      out[d] += acc + sigma * zs[j];
+    }
+  }
+}
+
+extern "C" void launch_clip_rowsum_noise(const float* g, const float* sqnorms, float* out,
+                                         float clip_bound, float sigma, uint64_t seed,
+                                         uint64_t offset, int64_t B, int64_t D, hipStream_t s) {
+  clip_rowsum_noise_kernel<<<grid_1d((D + 3) / 4), BLOCK, 0, s>>>(g, sqnorms, out, clip_bound,
+                                                                  sigma, seed, offset, B, D);
+}
+
 // ---------------------------------------------------------------------------
 // Streaming confusion counts (K14): per-class TP/FP/FN/TN from argmax preds.
 // out layout: int64 [C, 4] = (tp, fp, fn, tn). Single pass, LDS-staged

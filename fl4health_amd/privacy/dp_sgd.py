@@ -103,8 +103,16 @@ class DpSgdEngine:
             self.module._ghost.clear()
         for p in params:
             g = torch.zeros(p.numel(), dtype=torch.float32, device=device)
-            F.clip_rowsum_(p.grad_sample.reshape(batch, -1).float(), sqnorms, g, eff_bound)
-            self._finalize_grad(p, g, sigma, final_div)
+            gs = p.grad_sample.reshape(batch, -1).float()
+            if self.noise_multiplier > 0 and final_div == 1.0 and g.is_cuda:
+                # fused single pass: clipped per-sample sum + DP noise (K7
+                # epilogue fusion; stream-identical to the unfused pair)
+                F.clip_rowsum_noise_(gs, sqnorms, g, eff_bound, sigma, self.seed, self._noise_counter)
+                self._noise_counter += (p.numel() + 3) // 4 + 1
+                self._write_grad(p, g)
+            else:
+                F.clip_rowsum_(gs, sqnorms, g, eff_bound)
+                self._finalize_grad(p, g, sigma, final_div)
             p.grad_sample = None
         self.optimizer.step()
 
@@ -117,6 +125,10 @@ class DpSgdEngine:
             self._noise_counter += (p.numel() + 3) // 4 + 1
         if final_div != 1.0:
             g /= final_div
+        self._write_grad(p, g)
+
+    def _write_grad(self, p: torch.Tensor, g: torch.Tensor) -> None:
+        g = g.reshape(-1)
         if p.grad is not None and p.grad.shape == p.shape:
             p.grad.copy_(g.view(p.shape).to(p.grad.dtype))
         else:

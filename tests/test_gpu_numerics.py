@@ -527,3 +527,24 @@ def test_fused_convblock3d_trains():
         loss = sum(o.float().pow(2).mean() for o in out) if isinstance(out, list) else out.float().pow(2).mean()
     loss.backward()
     assert all(p.grad is not None for p in model.parameters() if p.requires_grad)
+
+
+@requires_gpu
+def test_clip_rowsum_noise_fused_matches_unfused():
+    """Fused clip+noise kernel produces EXACTLY the unfused pair's result
+    for the same Philox (seed, offset)."""
+    from fl4health_amd.ops import functional as F
+
+    torch.manual_seed(0)
+    b, d = 8, 1023  # odd D exercises the 4-element philox tail
+    g = torch.randn(b, d, device="cuda")
+    sq = (torch.randn(b, device="cuda") ** 2) * 5
+    out1 = torch.zeros(d, device="cuda")
+    F.clip_rowsum_(g, sq, out1, 0.7)
+    torch.cuda.synchronize()
+    from fl4health_amd import _C
+
+    _C.gaussian_noise_(out1, 1.0, 0.3, 123, 77)
+    out2 = torch.zeros(d, device="cuda")
+    F.clip_rowsum_noise_(g, sq, out2, 0.7, sigma=0.3, seed=123, offset=77)
+    assert torch.allclose(out1, out2, atol=1e-6), float((out1 - out2).abs().max())
