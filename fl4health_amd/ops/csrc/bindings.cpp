@@ -21,6 +21,8 @@ void launch_bernoulli_mask(const float*, const float*, float*, float*, uint64_t,
                            int64_t, hipStream_t);
 void launch_per_sample_sqnorm(const float*, float*, int64_t, int64_t, hipStream_t);
 void launch_clip_rowsum(const float*, const float*, float*, float, int64_t, int64_t, hipStream_t);
+void launch_clip_rowsum_noise(const float*, const float*, float*, float, float, uint64_t,
+                              uint64_t, int64_t, int64_t, hipStream_t);
 void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, int64_t,
                       hipStream_t);
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
