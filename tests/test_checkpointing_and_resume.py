@@ -106,3 +106,90 @@ def test_kill_and_resume_fault_tolerance(tmp_path):
     assert srv_b.current_round == 2
     # round 1 was NOT re-executed: exactly one entry per round
     assert sorted(set(rounds_executed)) == rounds_executed
+
+
+def test_best_metric_checkpointer_max_and_min(tmp_path):
+    import torch.nn as nn
+
+    from fl4health_amd.checkpointing.checkpointer import BestMetricTorchModuleCheckpointer
+
+    m1 = nn.Linear(2, 2)
+    m2 = nn.Linear(2, 2)
+    ck = BestMetricTorchModuleCheckpointer(str(tmp_path), "best_acc.pt", metric_name="val - accuracy", maximize=True)
+    ck.maybe_checkpoint(m1, 1.0, {"val - accuracy": 0.5})
+    first = {k: v.clone() for k, v in torch.load(tmp_path / "best_acc.pt", weights_only=False).state_dict().items()}
+    ck.maybe_checkpoint(m2, 1.0, {"val - accuracy": 0.4})  # worse: must NOT overwrite
+    again = torch.load(tmp_path / "best_acc.pt", weights_only=False).state_dict()
+    assert all(torch.equal(first[k], again[k]) for k in first)
+    ck.maybe_checkpoint(m2, 1.0, {"val - accuracy": 0.9})  # better: overwrites
+    best = torch.load(tmp_path / "best_acc.pt", weights_only=False).state_dict()
+    assert any(not torch.equal(first[k], best[k]) for k in first)
+
+
+def test_opacus_checkpointer_strips_wrapper(tmp_path):
+    import torch.nn as nn
+
+    from fl4health_amd.checkpointing.opacus_checkpointer import BestLossOpacusCheckpointer, LatestOpacusCheckpointer
+    from fl4health_amd.privacy.grad_sample import GradSampleModule
+
+    inner = nn.Sequential(nn.Linear(4, 4), nn.ReLU(), nn.Linear(4, 2))
+    gsm = GradSampleModule(inner)
+    ck = BestLossOpacusCheckpointer(str(tmp_path), "dp_best.pt")
+    ck.maybe_checkpoint(gsm, 0.5, {})
+    state = torch.load(tmp_path / "dp_best.pt", weights_only=False)
+    # saved keys are the UNWRAPPED module's keys (no _module. prefixes)
+    assert set(state.keys()) == set(inner.state_dict().keys())
+    fresh = nn.Sequential(nn.Linear(4, 4), nn.ReLU(), nn.Linear(4, 2))
+    ck.load_best_checkpoint_into_model(fresh)
+    assert torch.equal(fresh[0].weight, inner[0].weight)
+    latest = LatestOpacusCheckpointer(str(tmp_path), "dp_latest.pt")
+    latest.maybe_checkpoint(gsm, 9.9, {})
+    assert (tmp_path / "dp_latest.pt").exists()
+
+
+def test_packed_server_hydration_variants(tmp_path):
+    """Packed-format server checkpointing must strip aux payloads before
+    hydration (reference server_module.py:205-441): scaffold variates,
+    adaptive mu, clipping bit, layer names."""
+    from fl4health_amd.checkpointing.checkpointer import LatestTorchModuleCheckpointer
+    from fl4health_amd.checkpointing.server_module import (
+        AdaptiveConstraintServerCheckpointAndStateModule,
+        ClippingBitServerCheckpointAndStateModule,
+        ScaffoldServerCheckpointAndStateModule,
+    )
+    from fl4health_amd.common import Parameters
+    from fl4health_amd.parameter_exchange.exchangers import FullParameterExchangerWithPacking
+    from fl4health_amd.parameter_exchange.flat import FlatParameterView
+    from fl4health_amd.parameter_exchange.packers import (
+        ParameterPackerAdaptiveConstraint,
+        ParameterPackerWithClippingBit,
+        ParameterPackerWithControlVariates,
+    )
+
+    import torch.nn as nn
+
+    def mk_model():
+        # buffer-free model: integer buffers (BN num_batches_tracked) round
+        # through the fp32 flat and would fail exact comparison
+        torch.manual_seed(0)
+        return nn.Sequential(nn.Linear(8, 8), nn.ReLU(), nn.Linear(8, 3))
+
+    for module_cls, packer, aux in [
+        (ScaffoldServerCheckpointAndStateModule, ParameterPackerWithControlVariates(), torch.zeros(10)),
+        (AdaptiveConstraintServerCheckpointAndStateModule, ParameterPackerAdaptiveConstraint(), 0.5),
+        (ClippingBitServerCheckpointAndStateModule, ParameterPackerWithClippingBit(), 1.0),
+    ]:
+        model = mk_model()
+        view = FlatParameterView(model)
+        flat = view.flat.clone()
+        target = torch.randn_like(flat)
+        packed = packer.pack_parameters(Parameters([target.clone()]), aux)
+        mod = module_cls(
+            model=model,
+            parameter_exchanger=FullParameterExchangerWithPacking(packer),
+            model_checkpointers=LatestTorchModuleCheckpointer(str(tmp_path), f"{module_cls.__name__}.pt"),
+        )
+        mod.maybe_checkpoint(packed, 1.0, {})
+        hydrated = torch.load(tmp_path / f"{module_cls.__name__}.pt", weights_only=False)
+        hv = FlatParameterView(hydrated)
+        assert torch.allclose(hv.flat, target), module_cls.__name__
