@@ -66,9 +66,26 @@ class ResNet18(nn.Module):
         return self.fc(out)
 
 
+class _FusedBasicBlock(BasicBlock):
+    """BasicBlock whose epilogue (bn2 + residual add + relu) runs the fused
+    CDNA kernel: the residual add and final ReLU fold into bn2's normalize
+    pass, and the backward emits the residual's masked dy from the same
+    reduction kernel (no separate add/relu/relu-backward launches)."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from fl4health_amd.ops.batchnorm import CdnaBatchNorm2d
+
+        out = self.act1(self.bn1(self.conv1(x)))
+        out = self.conv2(out)
+        res = self.shortcut(x)
+        if isinstance(self.bn2, CdnaBatchNorm2d) and self.training:
+            return self.bn2.forward_add_relu(out, res)
+        return self.act2(self.bn2(out) + res)
+
+
 def fuse_resnet_bn_relu(model: "ResNet18") -> "ResNet18":
-    """Fuse bn->relu pairs into the CDNA BatchNorm kernel (the relu AFTER the
-    residual add — act2 — cannot be fused into bn2 and stays eager)."""
+    """Fuse bn->relu pairs into the CDNA BatchNorm kernel, and each basic
+    block's bn2 + residual-add + relu epilogue into ONE fused kernel."""
     from fl4health_amd.ops.batchnorm import CdnaBatchNorm2d, _FusedReluIdentity
 
     if isinstance(model.bn1, CdnaBatchNorm2d):
@@ -79,4 +96,6 @@ def fuse_resnet_bn_relu(model: "ResNet18") -> "ResNet18":
             if isinstance(block.bn1, CdnaBatchNorm2d):
                 block.bn1.fuse_relu = True
                 block.act1 = _FusedReluIdentity()
+            if type(block) is BasicBlock:
+                block.__class__ = _FusedBasicBlock
     return model

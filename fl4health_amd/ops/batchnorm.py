@@ -19,7 +19,7 @@ class _CdnaBatchNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu):
         F._require_ext("bn_fwd_train")
-        y, mean, invstd = F._C.bn_fwd_train(x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu)
+        y, mean, invstd = F._C.bn_fwd_train(x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu, None)
         ctx.save_for_backward(x2d, mean, invstd, gamma, beta)
         ctx.fuse_relu = fuse_relu
         return y
@@ -27,8 +27,31 @@ class _CdnaBatchNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x2d, mean, invstd, gamma, beta = ctx.saved_tensors
-        dx, dgamma, dbeta = F._C.bn_bwd(x2d, dy.contiguous(), mean, invstd, gamma, beta, ctx.fuse_relu)
+        dx, dgamma, dbeta = F._C.bn_bwd(x2d, dy.contiguous(), mean, invstd, gamma, beta, ctx.fuse_relu, None)
         return dx, dgamma, dbeta, None, None, None, None, None
+
+
+class _CdnaBatchNormAddReluFn(torch.autograd.Function):
+    """BN + residual add + ReLU in the normalize pass (the last eager
+    elementwise op of a ResNet basic block folded into the BN kernel); the
+    backward emits BOTH dx (through BN) and the residual's masked dy."""
+
+    @staticmethod
+    def forward(ctx, x2d, res2d, gamma, beta, running_mean, running_var, momentum, eps):
+        F._require_ext("bn_fwd_train")
+        y, mean, invstd = F._C.bn_fwd_train(
+            x2d, gamma, beta, running_mean, running_var, momentum, eps, True, res2d
+        )
+        ctx.save_for_backward(x2d, res2d, mean, invstd, gamma, beta)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, res2d, mean, invstd, gamma, beta = ctx.saved_tensors
+        dx, dgamma, dbeta, dres = F._C.bn_bwd(
+            x2d, dy.contiguous(), mean, invstd, gamma, beta, True, res2d
+        )
+        return dx, dres, dgamma, dbeta, None, None, None, None
 
 
 class CdnaBatchNorm2d(nn.BatchNorm2d):
@@ -61,6 +84,38 @@ class CdnaBatchNorm2d(nn.BatchNorm2d):
         y2d = _CdnaBatchNormFn.apply(
             x2d, self.weight.float(), self.bias.float(), self.running_mean, self.running_var,
             float(self.momentum), float(self.eps), self.fuse_relu,
+        )
+        return y2d.view(n, h, w, c).permute(0, 3, 1, 2)
+
+    def forward_add_relu(self, input: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
+        """relu(bn(input) + residual) with the add+relu fused into the
+        normalize kernel (ResNet basic-block epilogue)."""
+        use_custom = (
+            input.is_cuda
+            and self.training
+            and self.affine
+            and self.track_running_stats
+            and self.momentum is not None
+            and input.dim() == 4
+            and input.is_contiguous(memory_format=torch.channels_last)
+            and input.dtype in (torch.bfloat16, torch.float32)
+            and residual.dtype == input.dtype
+            and F.HAS_EXT
+        )
+        if not use_custom:
+            return torch.relu(super().forward(input) + residual)
+        n, c, h, w = input.shape
+        x2d = input.permute(0, 2, 3, 1).reshape(n * h * w, c)
+        res2d = (
+            residual.permute(0, 2, 3, 1).reshape(n * h * w, c)
+            if residual.is_contiguous(memory_format=torch.channels_last)
+            else residual.contiguous(memory_format=torch.channels_last).permute(0, 2, 3, 1).reshape(n * h * w, c)
+        )
+        if self.num_batches_tracked is not None:
+            self.num_batches_tracked.add_(1)
+        y2d = _CdnaBatchNormAddReluFn.apply(
+            x2d, res2d, self.weight.float(), self.bias.float(), self.running_mean,
+            self.running_var, float(self.momentum), float(self.eps),
         )
         return y2d.view(n, h, w, c).permute(0, 3, 1, 2)
 

@@ -27,7 +27,7 @@ void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, 
                       hipStream_t);
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
 void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
-                   float*, float, float, int64_t, int, int, int, int, hipStream_t);
+                   float*, float, float, int64_t, int, int, int, int, const void*, hipStream_t);
 void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int, int, int, int,
                              int, hipStream_t);
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
@@ -46,7 +46,7 @@ void launch_mkmmd_backward(const float*, const float*, const float*, float*, int
                            int64_t, int, hipStream_t);
 void launch_bn_bwd(const void*, const void*, void*, float*, const float*, const float*,
                    const float*, const float*, float*, float*, float*, float*, int64_t, int, int,
-                   int, int, hipStream_t);
+                   int, int, const void*, void*, hipStream_t);
 }
 
 namespace {
@@ -247,7 +247,8 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
                                         c10::optional<torch::Tensor> beta,
                                         c10::optional<torch::Tensor> running_mean,
                                         c10::optional<torch::Tensor> running_var, double momentum,
-                                        double eps, bool fuse_relu) {
+                                        double eps, bool fuse_relu,
+                                        c10::optional<torch::Tensor> res) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2, "x must be [R, C] contiguous");
   int64_t R = x.size(0);
   int C = (int)x.size(1);
@@ -265,13 +266,15 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
       beta.has_value() ? beta->data_ptr<float>() : nullptr,
       running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
       running_var.has_value() ? running_var->data_ptr<float>() : nullptr, (float)momentum,
-      (float)eps, R, C, G, dtype, fuse_relu ? 1 : 0, stream());
+      (float)eps, R, C, G, dtype, fuse_relu ? 1 : 0,
+      res.has_value() ? res->data_ptr() : nullptr, stream());
   return {y, mean, invstd};
 }
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor mean,
                                   torch::Tensor invstd, c10::optional<torch::Tensor> gamma,
-                                  c10::optional<torch::Tensor> beta, bool fuse_relu) {
+                                  c10::optional<torch::Tensor> beta, bool fuse_relu,
+                                  c10::optional<torch::Tensor> res) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous() && x.sizes() == dy.sizes());
   int64_t R = x.size(0);
   int C = (int)x.size(1);
@@ -284,12 +287,20 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tens
   auto sum_dy_xhat = torch::empty({C}, fopts);
   auto dgamma = torch::empty({C}, fopts);
   auto dbeta = torch::empty({C}, fopts);
+  torch::Tensor dres;
+  void* dres_ptr = nullptr;
+  if (res.has_value()) {
+    dres = torch::empty_like(x);
+    dres_ptr = dres.data_ptr();
+  }
   launch_bn_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), partial.data_ptr<float>(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
                 gamma.has_value() ? gamma->data_ptr<float>() : nullptr,
                 beta.has_value() ? beta->data_ptr<float>() : nullptr, sum_dy.data_ptr<float>(),
                 sum_dy_xhat.data_ptr<float>(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                R, C, G, dtype, fuse_relu ? 1 : 0, stream());
+                R, C, G, dtype, fuse_relu ? 1 : 0,
+                res.has_value() ? res->data_ptr() : nullptr, dres_ptr, stream());
+  if (res.has_value()) return {dx, dgamma, dbeta, dres};
   return {dx, dgamma, dbeta};
 }
 

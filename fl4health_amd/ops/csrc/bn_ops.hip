@@ -104,7 +104,7 @@ template <typename T>
 __global__ __launch_bounds__(BNBLOCK) void bn_fwd_norm_kernel(
     const T* __restrict__ x, T* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ beta, int fuse_relu, int64_t R, int C) {
+    const float* __restrict__ beta, const T* __restrict__ res, int fuse_relu, int64_t R, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float m = mean[c];
@@ -115,6 +115,7 @@ __global__ __launch_bounds__(BNBLOCK) void bn_fwd_norm_kernel(
   float shift = b - m * scale;
   for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
     float v = fmaf(ld<T>(x + r * C + c), scale, shift);
+    if (res != nullptr) v += ld<T>(res + r * C + c);  // fused residual add
     if (fuse_relu) v = fmaxf(v, 0.0f);
     st<T>(y + r * C + c, v);
   }
@@ -127,8 +128,8 @@ template <typename T>
 __global__ __launch_bounds__(BNBLOCK) void bn_bwd_reduce_kernel(
     const T* __restrict__ x, const T* __restrict__ dy, float* __restrict__ partial,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ gamma, const float* __restrict__ beta, int fuse_relu, int64_t R,
-    int C, int G) {
+    const float* __restrict__ gamma, const float* __restrict__ beta, const T* __restrict__ res,
+    T* __restrict__ dres, int fuse_relu, int64_t R, int C, int G) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   int g = blockIdx.y;
@@ -142,7 +143,11 @@ __global__ __launch_bounds__(BNBLOCK) void bn_bwd_reduce_kernel(
   for (int64_t r = g; r < R; r += G) {
     float gy = ld<T>(dy + r * C + c);
     float xh = (ld<T>(x + r * C + c) - m) * is;
-    if (fuse_relu && fmaf(xh, gm, bt) <= 0.0f) gy = 0.0f;
+    float pre = fmaf(xh, gm, bt);
+    if (res != nullptr) pre += ld<T>(res + r * C + c);
+    if (fuse_relu && pre <= 0.0f) gy = 0.0f;
+    // the residual branch receives exactly the relu-masked dy
+    if (dres != nullptr) st<T>(dres + r * C + c, gy);
     sdy += gy;
     sdyx = fmaf(gy, xh, sdyx);
   }
@@ -186,8 +191,8 @@ __global__ __launch_bounds__(BNBLOCK) void bn_bwd_dx_kernel(
     const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
-    const float* __restrict__ sum_dy, const float* __restrict__ sum_dy_xhat, int fuse_relu,
-    int64_t R, int C) {
+    const float* __restrict__ sum_dy, const float* __restrict__ sum_dy_xhat,
+    const T* __restrict__ res, int fuse_relu, int64_t R, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float m = mean[c];
@@ -200,7 +205,9 @@ __global__ __launch_bounds__(BNBLOCK) void bn_bwd_dx_kernel(
   for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
     float gy = ld<T>(dy + r * C + c);
     float xh = (ld<T>(x + r * C + c) - m) * is;
-    if (fuse_relu && fmaf(xh, g, bt) <= 0.0f) gy = 0.0f;
+    float pre = fmaf(xh, g, bt);
+    if (res != nullptr) pre += ld<T>(res + r * C + c);
+    if (fuse_relu && pre <= 0.0f) gy = 0.0f;
     st<T>(dx + r * C + c, k * (gy - mean_dy - xh * mean_dy_xhat));
   }
 }
@@ -220,7 +227,7 @@ static inline void bn_dims(int C, int64_t R, int G, dim3* grid, dim3* block) {
 extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mean, float* invstd,
                               const float* gamma, const float* beta, float* running_mean,
                               float* running_var, float momentum, float eps, int64_t R, int C,
-                              int G, int dtype, int fuse_relu, hipStream_t s) {
+                              int G, int dtype, int fuse_relu, const void* res, hipStream_t s) {
   dim3 grid, block;
   bn_dims(C, R, G, &grid, &block);
   if (dtype == 1) {
@@ -234,10 +241,12 @@ extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mea
   // normalize: reuse G-deep row grid (bandwidth-bound)
   if (dtype == 1) {
     bn_fwd_norm_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
-        (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, invstd, gamma, beta, fuse_relu, R, C);
+        (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, invstd, gamma, beta,
+        (const __hip_bfloat16*)res, fuse_relu, R, C);
   } else {
     bn_fwd_norm_kernel<float><<<grid, block, 0, s>>>((const float*)x, (float*)y, mean, invstd,
-                                                     gamma, beta, fuse_relu, R, C);
+                                                     gamma, beta, (const float*)res, fuse_relu,
+                                                     R, C);
   }
 }
 
@@ -245,27 +254,28 @@ extern "C" void launch_bn_bwd(const void* x, const void* dy, void* dx, float* pa
                               const float* mean, const float* invstd, const float* gamma,
                               const float* beta, float* sum_dy, float* sum_dy_xhat, float* dgamma,
                               float* dbeta, int64_t R, int C, int G, int dtype, int fuse_relu,
-                              hipStream_t s) {
+                              const void* res, void* dres, hipStream_t s) {
   dim3 grid, block;
   bn_dims(C, R, G, &grid, &block);
   if (dtype == 1) {
     bn_bwd_reduce_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, partial, mean, invstd, gamma, beta,
-        fuse_relu, R, C, G);
+        (const __hip_bfloat16*)res, (__hip_bfloat16*)dres, fuse_relu, R, C, G);
   } else {
     bn_bwd_reduce_kernel<float><<<grid, block, 0, s>>>((const float*)x, (const float*)dy, partial,
-                                                       mean, invstd, gamma, beta, fuse_relu, R, C,
-                                                       G);
+                                                       mean, invstd, gamma, beta,
+                                                       (const float*)res, (float*)dres, fuse_relu,
+                                                       R, C, G);
   }
   bn_bwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
       partial, sum_dy, sum_dy_xhat, dgamma, dbeta, C, G);
   if (dtype == 1) {
     bn_bwd_dx_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, (__hip_bfloat16*)dx, mean, invstd,
-        gamma, beta, sum_dy, sum_dy_xhat, fuse_relu, R, C);
+        gamma, beta, sum_dy, sum_dy_xhat, (const __hip_bfloat16*)res, fuse_relu, R, C);
   } else {
     bn_bwd_dx_kernel<float><<<grid, block, 0, s>>>((const float*)x, (const float*)dy, (float*)dx,
                                                    mean, invstd, gamma, beta, sum_dy, sum_dy_xhat,
-                                                   fuse_relu, R, C);
+                                                   (const float*)res, fuse_relu, R, C);
   }
 }

@@ -225,8 +225,8 @@ def test_cdna_batchnorm_deterministic():
     from fl4health_amd.ops import functional as F
 
     x = torch.randn(100000, 64, device="cuda")
-    y1, m1, v1 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5, False)
-    y2, m2, v2 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5, False)
+    y1, m1, v1 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5, False, None)
+    y2, m2, v2 = F._C.bn_fwd_train(x, torch.ones(64, device="cuda"), torch.zeros(64, device="cuda"), None, None, 0.1, 1e-5, False, None)
     assert torch.equal(m1, m2) and torch.equal(v1, v2) and torch.equal(y1, y2)
 
 
@@ -548,3 +548,46 @@ def test_clip_rowsum_noise_fused_matches_unfused():
     out2 = torch.zeros(d, device="cuda")
     F.clip_rowsum_noise_(g, sq, out2, 0.7, sigma=0.3, seed=123, offset=77)
     assert torch.allclose(out1, out2, atol=1e-6), float((out1 - out2).abs().max())
+
+
+@requires_gpu
+def test_bn_add_relu_fused_matches_oracle():
+    """Fused BN + residual add + ReLU (the ResNet basic-block epilogue) vs
+    the eager oracle: y, dx, d(residual), dgamma, dbeta."""
+    from fl4health_amd.ops.batchnorm import _CdnaBatchNormAddReluFn
+
+    torch.manual_seed(0)
+    for dtype in (torch.float32, torch.bfloat16):
+        r, c = 4096, 64
+        base_x = torch.randn(r, c, device="cuda")
+        base_res = torch.randn(r, c, device="cuda")
+        x16 = base_x.to(dtype).detach()
+        res16 = base_res.to(dtype).detach()
+        x1 = x16.float().detach().clone().requires_grad_(True)
+        res1 = res16.float().detach().clone().requires_grad_(True)
+        g1 = torch.rand(c, device="cuda").requires_grad_(True)
+        b1 = torch.randn(c, device="cuda").requires_grad_(True)
+        rm = torch.zeros(c, device="cuda")
+        rv = torch.ones(c, device="cuda")
+        ref = torch.relu(
+            torch.nn.functional.batch_norm(x1, None, None, g1, b1, training=True, eps=1e-5) + res1
+        )
+        x2 = x16.detach().clone().requires_grad_(True)
+        res2 = res16.detach().clone().requires_grad_(True)
+        g2 = g1.detach().clone().requires_grad_(True)
+        b2 = b1.detach().clone().requires_grad_(True)
+        out = _CdnaBatchNormAddReluFn.apply(x2, res2, g2, b2, rm.clone(), rv.clone(), 0.1, 1e-5)
+        rel = (out.float() - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
+        assert rel < (1e-4 if dtype == torch.float32 else 3e-2), (dtype, float(rel))
+        gy = torch.randn_like(ref)
+        ref.backward(gy)
+        out.backward(gy.to(dtype))
+        tol = 1e-3 if dtype == torch.float32 else 6e-2
+        for name, (a, bb) in {
+            "dx": (x1.grad, x2.grad.float()),
+            "dres": (res1.grad, res2.grad.float()),
+            "dgamma": (g1.grad, g2.grad),
+            "dbeta": (b1.grad, b2.grad),
+        }.items():
+            rr = (a - bb).abs().max() / a.abs().max().clamp(min=1e-5)
+            assert rr < tol, (dtype, name, float(rr))

@@ -14,8 +14,8 @@ for R, C in [(131072, 64), (32768, 128), (8192, 256), (2048, 512)]:
     dy = torch.randn_like(x)
     gamma = torch.ones(C, device="cuda"); beta = torch.zeros(C, device="cuda")
     rm = torch.zeros(C, device="cuda"); rv = torch.ones(C, device="cuda")
-    ms_f = t(lambda: F._C.bn_fwd_train(x, gamma, beta, rm, rv, 0.1, 1e-5))
-    y, mean, invstd = F._C.bn_fwd_train(x, gamma, beta, rm, rv, 0.1, 1e-5)
+    ms_f = t(lambda: F._C.bn_fwd_train(x, gamma, beta, rm, rv, 0.1, 1e-5, False, None))
+    y, mean, invstd = F._C.bn_fwd_train(x, gamma, beta, rm, rv, 0.1, 1e-5, False, None)
     ms_b = t(lambda: F._C.bn_bwd(x, dy, mean, invstd, gamma))
     bytes_f = R * C * 2 * 3  # read x twice + write y
     bytes_b = R * C * 2 * 5
