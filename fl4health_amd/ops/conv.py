@@ -137,14 +137,13 @@ class CdnaConv2d(nn.Conv2d):
     kernel on GPU bf16 channels-last inputs; falls back to F.conv2d off that
     fast path (CPU, fp32 eval, unsupported widths, other hyper-params).
 
-    `mfma_widths` gates adoption to the spatial widths where a kernel variant
-    MEASURES faster than TUNED MIOpen (cudnn.benchmark find): the KB=32
+    Adoption is gated per shape to where a kernel variant MEASURES faster
+    than TUNED MIOpen (cudnn.benchmark find; see _use_kb32): the KB=32
     fully-pipelined kernel at 4x4 C512 (1.23x) and 8x8 C<=128 (1.50x). The
     other variants (8-wave persistent-weight glds at 32x32, round-1 direct
     at 16x16) beat MIOpen's *default* solver picks by 1.2-1.9x but lose to
     its tuned picks and stay opt-in via `force_mfma`."""
 
-    mfma_widths: frozenset = frozenset({4, 8})
     force_mfma: bool = False  # True: run our kernels on every supported width
 
     def _fast_path(self, input: torch.Tensor) -> bool:
