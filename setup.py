@@ -29,9 +29,15 @@ setup(
                 os.path.join(CSRC, "in_ops.hip"),
             ],
             extra_compile_args={
-                "cxx": ["-O3"],
+                # FL4_ASAN=1: host-side AddressSanitizer build of the binding
+                # layer (run pytest with LD_PRELOAD=$(gcc -print-file-name=libasan.so)
+                # ASAN_OPTIONS=detect_leaks=0). Device-side ASAN on gfx950
+                # needs an XNACK-enabled driver stack and is not wired here.
+                "cxx": ["-O3"] + (["-fsanitize=address", "-fno-omit-frame-pointer"]
+                                  if os.environ.get("FL4_ASAN") == "1" else []),
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
             },
+            extra_link_args=(["-fsanitize=address"] if os.environ.get("FL4_ASAN") == "1" else []),
         )
     ],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
