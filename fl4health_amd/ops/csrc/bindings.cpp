@@ -31,6 +31,9 @@ void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, con
 void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int, int, int, int,
                              int, hipStream_t);
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
+void launch_coo_count(const float*, float, int64_t, int64_t, int32_t*, hipStream_t);
+void launch_coo_write(const float*, const float*, float, const int32_t*, float*, int64_t*,
+                      const int64_t*, int, int64_t, int64_t, int64_t, hipStream_t);
 void launch_moon_contrastive(const float*, const float*, const float*, int, int, int64_t, float,
                              float*, float*, hipStream_t);
 void launch_in3d_fwd(const void*, void*, float*, float*, float*, const float*, const float*, int,
@@ -358,6 +361,38 @@ torch::Tensor pack_kb32(torch::Tensor w, bool bwd) {
   return out;
 }
 
+// K11: deterministic GPU stream compaction — (values[score >= t],
+// per-dim COO indices) in row-major order, matching nonzero()'s ordering.
+std::vector<torch::Tensor> coo_compact(torch::Tensor values, torch::Tensor score,
+                                       double threshold) {
+  check_f32(values, "values");
+  check_f32(score, "score");
+  TORCH_CHECK(values.sizes() == score.sizes(), "values/score shape mismatch");
+  int ndim = (int)values.dim();
+  TORCH_CHECK(ndim >= 1 && ndim <= 8, "1..8 dims supported");
+  int64_t n = values.numel();
+  int64_t n_chunks = (n + 63) / 64;
+  auto iopts = torch::TensorOptions().dtype(torch::kInt32).device(values.device());
+  auto counts = torch::empty({n_chunks}, iopts);
+  launch_coo_count(score.data_ptr<float>(), (float)threshold, n, n_chunks,
+                   counts.data_ptr<int>(), stream());
+  auto csum = counts.cumsum(0, torch::kInt32);
+  auto offsets = csum - counts;  // exclusive
+  int64_t nnz = csum.numel() ? csum[-1].item<int64_t>() : 0;  // ONE small sync
+  auto values_out = torch::empty({nnz}, values.options());
+  auto lopts = torch::TensorOptions().dtype(torch::kInt64).device(values.device());
+  auto indices_out = torch::empty({ndim, nnz}, lopts);
+  auto dims = torch::tensor(values.sizes().vec(), torch::TensorOptions().dtype(torch::kInt64))
+                  .to(values.device());
+  if (nnz > 0) {
+    launch_coo_write(values.data_ptr<float>(), score.data_ptr<float>(), (float)threshold,
+                     offsets.to(torch::kInt32).data_ptr<int>(), values_out.data_ptr<float>(),
+                     indices_out.data_ptr<int64_t>(), dims.data_ptr<int64_t>(), ndim, n,
+                     n_chunks, nnz, stream());
+  }
+  return {values_out, indices_out};
+}
+
 // Fused MOON contrastive loss + input gradient (K8): one workgroup per
 // sample computes the cosine-similarity logits against [pos | negs], the
 // softmax-CE loss, and dz in a single launch (partners are frozen snapshots).
@@ -489,6 +524,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_fwd_kb32", &conv3x3_fwd_kb32,
         "direct 3x3 NHWC bf16 conv forward, KB=32 glds-pipelined variant");
   m.def("pack_kb32", &pack_kb32, "fused weight pack for conv3x3_fwd_kb32");
+  m.def("coo_compact", &coo_compact,
+        "deterministic stream compaction: values + per-dim COO indices above a threshold");
   m.def("moon_contrastive", &moon_contrastive,
         "fused MOON contrastive loss + dz (cosine logits, softmax-CE, label 0)");
   m.def("in3d_fwd", &in3d_fwd, "fused InstanceNorm3d + LeakyReLU forward (NCDHW bf16)");

@@ -538,3 +538,62 @@ extern "C" void launch_weighted_sum_rows(const float* stack, const float* w, flo
                                          int64_t n, hipStream_t s) {
   weighted_sum_rows_kernel<<<grid_1d(n), BLOCK, 0, s>>>(stack, w, out, K, n);
 }
+
+// ---------------------------------------------------------------------------
+// K11: deterministic stream compaction for sparse COO packing (reference
+// parameter_packer.py:94-142 + sparse_coo_parameter_exchanger.py:18 do this
+// with host-side nonzero/scatter). Two passes around ONE torch cumsum:
+//   1) per-thread-chunk hit counts (each thread owns a contiguous chunk, so
+//      the final ordering is row-major ascending — identical to nonzero())
+//   2) each thread writes its chunk's selected values + per-dim indices at
+//      its exclusive offset.
+// ---------------------------------------------------------------------------
+#define COMPACT_CHUNK 64
+
+__global__ __launch_bounds__(BLOCK) void coo_count_kernel(
+    const float* __restrict__ score, float threshold, int64_t n, int64_t n_chunks,
+    int32_t* __restrict__ counts) {
+  GSL(c, n_chunks, STRIDE) {
+    int64_t lo = c * COMPACT_CHUNK;
+    int64_t hi = min(lo + COMPACT_CHUNK, n);
+    int cnt = 0;
+    for (int64_t i = lo; i < hi; ++i) cnt += (score[i] >= threshold) ? 1 : 0;
+    counts[c] = cnt;
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void coo_write_kernel(
+    const float* __restrict__ values_in, const float* __restrict__ score, float threshold,
+    const int32_t* __restrict__ offsets,  // exclusive per-chunk offsets
+    float* __restrict__ values_out, int64_t* __restrict__ indices_out,  // [ndim, nnz]
+    const int64_t* __restrict__ dims, int ndim, int64_t n, int64_t n_chunks, int64_t nnz) {
+  GSL(c, n_chunks, STRIDE) {
+    int64_t lo = c * COMPACT_CHUNK;
+    int64_t hi = min(lo + COMPACT_CHUNK, n);
+    int64_t out = offsets[c];
+    for (int64_t i = lo; i < hi; ++i) {
+      if (score[i] < threshold) continue;
+      values_out[out] = values_in[i];
+      int64_t rem = i;
+      for (int d = ndim - 1; d >= 0; --d) {
+        indices_out[(int64_t)d * nnz + out] = rem % dims[d];
+        rem /= dims[d];
+      }
+      ++out;
+    }
+  }
+}
+
+extern "C" void launch_coo_count(const float* score, float threshold, int64_t n,
+                                 int64_t n_chunks, int32_t* counts, hipStream_t s) {
+  coo_count_kernel<<<grid_1d(n_chunks), BLOCK, 0, s>>>(score, threshold, n, n_chunks, counts);
+}
+
+extern "C" void launch_coo_write(const float* values_in, const float* score, float threshold,
+                                 const int32_t* offsets, float* values_out, int64_t* indices_out,
+                                 const int64_t* dims, int ndim, int64_t n, int64_t n_chunks,
+                                 int64_t nnz, hipStream_t s) {
+  coo_write_kernel<<<grid_1d(n_chunks), BLOCK, 0, s>>>(values_in, score, threshold, offsets,
+                                                       values_out, indices_out, dims, ndim, n,
+                                                       n_chunks, nnz);
+}

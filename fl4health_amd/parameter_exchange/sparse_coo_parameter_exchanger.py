@@ -29,10 +29,26 @@ class SparseCooParameterExchanger(ParameterExchanger):
         threshold = torch.kthvalue(all_scores.float().cpu(), all_scores.numel() - k + 1).values
         sd = model.state_dict()
         values, indices, shapes, names = [], [], [], []
+        use_kernel = all(s.is_cuda for s in scores.values())
+        if use_kernel:
+            try:
+                from fl4health_amd import _C
+            except ImportError:
+                use_kernel = False
         for name, score in scores.items():
-            mask = score >= threshold.to(score.device)
-            nz = mask.nonzero().t()
-            values.append(sd[name][mask].detach().reshape(-1).float())
+            if use_kernel:
+                # K11: one-kernel deterministic stream compaction (row-major
+                # order identical to nonzero()) instead of mask/nonzero/
+                # boolean-index host round-trips
+                vals, nz = _C.coo_compact(
+                    sd[name].detach().float().contiguous(), score.float().contiguous(),
+                    float(threshold),
+                )
+            else:
+                mask = score >= threshold.to(score.device)
+                nz = mask.nonzero().t()
+                vals = sd[name][mask].detach().reshape(-1).float()
+            values.append(vals)
             indices.append(nz)
             shapes.append(list(score.shape))
             names.append(name)
