@@ -84,3 +84,15 @@ def test_new_research_harnesses_run(module, extra):
     )
     assert out.returncode == 0, out.stderr[-2000:]
     assert '"algorithm"' in out.stdout
+
+
+@pytest.mark.parametrize("alg", ["apfl", "scaffold", "local"])
+def test_gemini_harness_runs(alg):
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [sys.executable, "-m", "research.gemini.run_experiment", "--algorithm", alg,
+         "--n_clients", "2", "--rounds", "1", "--local_steps", "1", "--batch_size", "16"],
+        capture_output=True, text=True, timeout=300, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert '"algorithm"' in out.stdout
