@@ -222,3 +222,29 @@ def test_cdna_conv_class_swap_and_fallback_cpu():
     x = torch.randn(2, 3, 8, 8)
     # CPU path falls back to F.conv2d exactly
     assert torch.allclose(model(x), ref(x), atol=1e-6)
+
+
+def test_conv3d_depth_decomposition_matches_conv3d_cpu():
+    """Cdna3dConv's 3-tap depth decomposition == F.conv3d (fwd + dx + dw);
+    the variant is opt-in (loses to tuned 3D solvers on GPU — negative
+    result recorded in profiles/unet3d_round2.md) but must stay correct."""
+    import torch.nn as nn
+
+    from fl4health_amd.ops.conv3d import Cdna3dConv, _conv3x3x3_by_2d, convert_conv3d_to_cdna
+
+    torch.manual_seed(0)
+    n, c, k, d, h, w = 2, 3, 5, 4, 6, 5
+    conv = nn.Conv3d(c, k, 3, padding=1)
+    x1 = torch.randn(n, c, d, h, w, requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    ref = conv(x1)
+    out = _conv3x3x3_by_2d(x2, conv.weight, conv.bias)
+    assert torch.allclose(out, ref, atol=1e-5)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    out.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, atol=1e-5)
+    m = convert_conv3d_to_cdna(nn.Sequential(nn.Conv3d(2, 2, 3, padding=1)))
+    assert type(m[0]) is Cdna3dConv
+    y = m(torch.randn(1, 2, 4, 4, 4))  # CPU falls back to F.conv3d
+    assert y.shape == (1, 2, 4, 4, 4)
