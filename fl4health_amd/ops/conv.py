@@ -54,22 +54,43 @@ def _image_kb32(w9: torch.Tensor) -> torch.Tensor:
     return torch.where(kk_mask, swapped, img)
 
 
-def _use_kb32(width: int, c: int, k: int) -> bool:
+def _variant_for(width: int, c: int, k: int) -> str | None:
     """Per-shape variant gate from the measured matrix vs TUNED MIOpen
     (cudnn.benchmark find, what training runs use — see
-    profiles/kernels_summary.md): the KB=32 fully-pipelined kernel wins at
-    4x4 C512 (1.23x) and 8x8 C<=128 (1.50x); tuned MIOpen wins the rest
-    (it reaches 300-360 TF at 16x16/32x32 where our kernels sit at
-    190-290)."""
+    profiles/kernels_summary.md):
+
+    - kb32 (both operands glds-pipelined) wins at 4x4 C512 (1.1-1.2x) and
+      8x8 C<=128 (1.5x)
+    - kzloop (input-resident multi-kz) wins at the asymmetric 16x16 shapes
+      (C128->K64 1.23x, C64->K128 1.21x) and is near-parity (0.94x) at
+      16x16 C128->K128
+    - tuned MIOpen keeps the rest (345-476 TF at the square 16x16/32x32
+      shapes)
+    """
     if c % 64 != 0 or k % 32 != 0:
-        return False
-    return width <= 4 or (width == 8 and c <= 128)
+        return None
+    if width <= 4 or (width == 8 and c <= 128):
+        return "kb32"
+    if width == 16 and c <= 128 and k <= 128 and c * k <= 128 * 64 and k >= 64:
+        return "kzloop"
+    return None
+
+
+def _use_kb32(width: int, c: int, k: int) -> bool:
+    return _variant_for(width, c, k) is not None
+
+
+def _run_variant(variant: str, x_nhwc: torch.Tensor, wimg: torch.Tensor, bias):
+    if variant == "kzloop":
+        return _C.conv3x3_fwd_kzloop(x_nhwc, wimg, bias)
+    return _C.conv3x3_fwd_kb32(x_nhwc, wimg, bias)
 
 
 def _run_fwd(x_nhwc: torch.Tensor, w9: torch.Tensor, bias: torch.Tensor | None) -> torch.Tensor:
     width, c, k = x_nhwc.shape[2], w9.shape[1], w9.shape[2]
-    if _use_kb32(width, c, k):
-        return _C.conv3x3_fwd_kb32(x_nhwc, _image_kb32(w9), bias)
+    variant = _variant_for(width, c, k)
+    if variant is not None:
+        return _run_variant(variant, x_nhwc, _image_kb32(w9), bias)
     return _C.conv3x3_fwd(x_nhwc, w9, bias)
 
 
@@ -77,10 +98,11 @@ class _CdnaConv3x3Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x_nhwc: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor | None):
         w16 = weight.to(torch.bfloat16)
-        if x_nhwc.is_cuda and _use_kb32(x_nhwc.shape[2], w16.shape[1], w16.shape[0]):
+        variant = _variant_for(x_nhwc.shape[2], w16.shape[1], w16.shape[0]) if x_nhwc.is_cuda else None
+        if variant is not None:
             # fused single-kernel pack straight to the swizzled LDS image
-            y = _C.conv3x3_fwd_kb32(
-                x_nhwc, _C.pack_kb32(w16.contiguous(), False),
+            y = _run_variant(
+                variant, x_nhwc, _C.pack_kb32(w16.contiguous(), False),
                 bias.float() if bias is not None else None,
             )
         else:
@@ -96,8 +118,9 @@ class _CdnaConv3x3Fn(torch.autograd.Function):
         n, h, w, k = gy.shape
         c = x_nhwc.shape[3]
         # dx: same kernel, rotated weights (conv roles swap: C_conv = K)
-        if gy.is_cuda and _use_kb32(gy.shape[2], k, c):
-            dx = _C.conv3x3_fwd_kb32(gy, _C.pack_kb32(w16.contiguous(), True), None)
+        bwd_variant = _variant_for(gy.shape[2], k, c) if gy.is_cuda else None
+        if bwd_variant is not None:
+            dx = _run_variant(bwd_variant, gy, _C.pack_kb32(w16.contiguous(), True), None)
         else:
             dx = _run_fwd(gy, _pack_bwd(w16), None)
         # dW: route to MIOpen's tuned wrw igemm. The round-1 "9 shifted

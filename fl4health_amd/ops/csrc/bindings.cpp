@@ -31,6 +31,8 @@ void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, con
 void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int, int, int, int,
                              int, hipStream_t);
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
+void launch_conv3x3_fwd_kzloop(const void*, const void*, const float*, void*, int, int, int, int,
+                               int, hipStream_t);
 void launch_coo_count(const float*, float, int64_t, int64_t, int32_t*, hipStream_t);
 void launch_coo_write(const float*, const float*, float, const int32_t*, float*, int64_t*,
                       const int64_t*, int, int64_t, int64_t, int64_t, hipStream_t);
@@ -347,6 +349,24 @@ torch::Tensor conv3x3_fwd_kb32(torch::Tensor x, torch::Tensor wimg,
   return y;
 }
 
+// Variant D: input-resident multi-kz (one block per tile loops every
+// K-block; input staged once, weights pipelined). Same wimg pack as kb32.
+torch::Tensor conv3x3_fwd_kzloop(torch::Tensor x, torch::Tensor wimg,
+                                 c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.dim() == 4 && x.is_contiguous());
+  TORCH_CHECK(wimg.is_cuda() && wimg.scalar_type() == torch::kBFloat16 && wimg.dim() == 5 &&
+              wimg.is_contiguous());
+  int64_t N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int64_t K = wimg.size(0) * 32;
+  TORCH_CHECK(wimg.size(1) * 64 == C && wimg.size(2) == 9, "wimg must be [K/32, C/64, 9, 32, 64]");
+  if (bias.has_value()) check_f32(*bias, "bias");
+  auto y = torch::empty({N, H, W, K}, x.options());
+  launch_conv3x3_fwd_kzloop(x.data_ptr(), wimg.data_ptr(),
+                            bias.has_value() ? bias->data_ptr<float>() : nullptr, y.data_ptr(),
+                            (int)N, (int)H, (int)W, (int)C, (int)K, stream());
+  return y;
+}
+
 // Fused weight pack for conv3x3_fwd_kb32: [K, C, 3, 3] -> swizzled LDS-image
 // slabs in ONE kernel (replaces a ~6-op torch chain per conv per step).
 // bwd=true packs the bwd-data weights (roles swapped, taps rotated).
@@ -524,6 +544,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_fwd_kb32", &conv3x3_fwd_kb32,
         "direct 3x3 NHWC bf16 conv forward, KB=32 glds-pipelined variant");
   m.def("pack_kb32", &pack_kb32, "fused weight pack for conv3x3_fwd_kb32");
+  m.def("conv3x3_fwd_kzloop", &conv3x3_fwd_kzloop,
+        "direct 3x3 conv, input-resident multi-kz variant (KB=32)");
   m.def("coo_compact", &coo_compact,
         "deterministic stream compaction: values + per-dim COO indices above a threshold");
   m.def("moon_contrastive", &moon_contrastive,

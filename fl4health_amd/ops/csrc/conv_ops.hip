@@ -575,3 +575,125 @@ extern "C" void launch_pack_kb32(const void* w, void* out, int K, int C, int mod
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   pack_kb32_kernel<<<blocks, 256, 0, s>>>((const bf16*)w, (bf16*)out, K, C, mode);
 }
+
+// ---------------------------------------------------------------------------
+// Variant D: input-resident multi-kz KB=32 ("kzloop"). For mid shapes
+// (16x16 C64/C128) the ENTIRE input halo fits LDS alongside two weight
+// slabs, so one block stages its input ONCE and loops every K-block with
+// the weight DMA pipelined — grid = n_tiles (fills at 256 tiles), no input
+// restaging per kz (variant C re-reads input KZ times), prologue amortized
+// over KZ*CC chunks.
+// ---------------------------------------------------------------------------
+#define CONVD_IN_CAP 1664      // 16-B chunks per input c-chunk (16x16: 1440, 8x8 SB2: 1600) + ragged-glds slack
+#define CONVD_MAX_CC 2         // resident input c-chunks
+__shared__ __bf16 s_convd[CONVD_MAX_CC * CONVD_IN_CAP * 8 + 2 * CONVC_WCHUNKS * 8];
+
+__global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kzloop_kernel(
+    const bf16* __restrict__ x,
+    const bf16* __restrict__ wimg,  // [K/32][C/64][9][32][64] contiguous slabs
+    const float* __restrict__ bias, bf16* __restrict__ y, int Nn, int H, int W, int C, int K,
+    int BH, int SB, int n_tiles, int h_groups) {
+  int tile = blockIdx.x;
+  int tile_w = W + 2;
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int n_cchunks = (C + CONV_CB - 1) / CONV_CB;
+  int kz = K / CONVC_KB;
+
+  int hg = tile % h_groups;
+  int n0 = (tile / h_groups) * SB;
+  int sb = min(SB, Nn - n0);
+  int h0 = hg * BH;
+  int bh = min(BH, H - h0);
+  int pps = bh * W;
+  int m_count = sb * pps;
+
+  __bf16* s_w0 = s_convd + CONVD_MAX_CC * CONVD_IN_CAP * 8;
+  // stage ALL input c-chunks once + the first weight slab
+  for (int ci = 0; ci < n_cchunks; ++ci)
+    issue_glds_input(x, s_convd + (size_t)ci * (CONVD_IN_CAP * 8), n0, h0, sb, bh, H, W, C,
+                     ci * CONV_CB, tile_w, tid, CONV_THREADS);
+  issue_glds_slab(wimg, s_w0, CONVC_WCHUNKS, tid);
+  __syncthreads();
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int q = 0; q < 2; ++q) acc[t][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  int total = kz * n_cchunks;
+  int cur = 0;
+  for (int it = 0; it < total; ++it) {
+    int kzi = it / n_cchunks;
+    int ci = it % n_cchunks;
+    if (it + 1 < total)
+      issue_glds_slab(wimg + (size_t)(it + 1) * (CONVC_WCHUNKS * 8),
+                      s_w0 + (size_t)(1 - cur) * (CONVC_WCHUNKS * 8), CONVC_WCHUNKS, tid);
+    const __bf16* sin = s_convd + (size_t)ci * (CONVD_IN_CAP * 8);
+    const __bf16* sw = s_w0 + (size_t)cur * (CONVC_WCHUNKS * 8);
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      int dy = tap / 3, dx = tap % 3;
+#pragma unroll
+      for (int ck = 0; ck < CONV_CB / 32; ++ck) {
+        int kbase = ck * 32 + (lane >> 4) * 8;
+        bf16x8 afrag[2];
+#pragma unroll
+        for (int t = 0; t < 2; ++t) {
+          int m = wave * 32 + t * 16 + (lane & 15);
+          int sidx = m / pps, rem = m % pps;
+          int hh = rem / W, ww = rem % W;
+          int apos = (sidx * (bh + 2) + hh + dy) * tile_w + (ww + dx);
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &sin[apos * CONV_CB + (kbase ^ conv_swz(apos))]);
+          if (m >= m_count) a = bf16x8{};
+          afrag[t] = a;
+        }
+#pragma unroll
+        for (int q = 0; q < 2; ++q) {
+          int br = tap * CONVC_KB + q * 16 + (lane & 15);
+          bf16x8 bfrag =
+              *reinterpret_cast<const bf16x8*>(&sw[br * CONV_CB + (kbase ^ conv_swz(br))]);
+#pragma unroll
+          for (int t = 0; t < 2; ++t)
+            acc[t][q] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[t], bfrag, acc[t][q], 0, 0, 0);
+        }
+      }
+    }
+    if (ci == n_cchunks - 1) {
+      int kb0 = kzi * CONVC_KB;
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int q = 0; q < 2; ++q) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            int m = wave * 32 + t * 16 + (lane >> 4) * 4 + r;
+            int k = q * 16 + (lane & 15);
+            if (m < m_count && kb0 + k < K) {
+              int sidx = m / pps, rem = m % pps;
+              int hh = rem / W, ww = rem % W;
+              float v = acc[t][q][r];
+              if (bias != nullptr) v += bias[kb0 + k];
+              y[(((int64_t)(n0 + sidx) * H + h0 + hh) * W + ww) * (int64_t)K + kb0 + k] = (bf16)v;
+            }
+          }
+          acc[t][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+        }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+}
+
+extern "C" void launch_conv3x3_fwd_kzloop(const void* x, const void* wimg, const float* bias,
+                                          void* y, int Nn, int H, int W, int C, int K,
+                                          hipStream_t s) {
+  int BH, SB, h_groups, n_tiles;
+  conv_tile_geom(Nn, H, W, 128, BH, SB, h_groups, n_tiles);
+  dim3 grid(n_tiles, 1, 1);
+  conv3x3_fwd_kzloop_kernel<<<grid, CONV_THREADS, 0, s>>>(
+      (const bf16*)x, (const bf16*)wimg, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
+      h_groups);
+}

@@ -75,6 +75,17 @@ def run_shape(n, h, w, c, k, iters=100):
         ok &= ok2
         t2 = bench(lambda: _C.conv3x3_fwd_kb32(x_nhwc, wimg, None), iters)
         line += f" | kb32 {t2:.3f} ms ({flops / t2 / 1e9:.0f} TF) {t_miopen / t2:.2f}x{'' if ok2 else ' FAIL rel=%.4f' % rel2}"
+        # variant D gate: all input chunks + 2 weight slabs resident
+        bh = min(max(128 // w, 1), h)
+        sb = max(128 // (h * w), 1)
+        in_chunks = sb * (bh + 2) * (w + 2) * 8
+        if c // 64 <= 2 and in_chunks + 63 <= 1664 and k >= 64:
+            out3 = _C.conv3x3_fwd_kzloop(x_nhwc, wimg, None).permute(0, 3, 1, 2)
+            rel3 = float(((out3.float() - ref.float()).abs().max() / refmax))
+            ok3 = rel3 < 2e-2
+            ok &= ok3
+            t3 = bench(lambda: _C.conv3x3_fwd_kzloop(x_nhwc, wimg, None), iters)
+            line += f" | kzloop {t3:.3f} ms ({flops / t3 / 1e9:.0f} TF) {t_miopen / t3:.2f}x{'' if ok3 else ' FAIL rel=%.4f' % rel3}"
 
     print(line)
     return ok
