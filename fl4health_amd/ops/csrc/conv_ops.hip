@@ -588,15 +588,24 @@ extern "C" void launch_pack_kb32(const void* w, void* out, int K, int C, int mod
 #define CONVD_MAX_CC 2         // resident input c-chunks
 __shared__ __bf16 s_convd[CONVD_MAX_CC * CONVD_IN_CAP * 8 + 2 * CONVC_WCHUNKS * 8];
 
-__global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kzloop_kernel(
+template <int WAVES>
+__global__ __launch_bounds__(WAVES * 64) void conv3x3_fwd_kzloop_kernel(
     const bf16* __restrict__ x,
     const bf16* __restrict__ wimg,  // [K/32][C/64][9][32][64] contiguous slabs
     const float* __restrict__ bias, bf16* __restrict__ y, int Nn, int H, int W, int C, int K,
     int BH, int SB, int n_tiles, int h_groups) {
+  // WAVES == 4: each wave owns a 32(M) x 32(K) tile (acc[2][2]).
+  // WAVES == 8: wave pairs split the K dim — wave (w&3) covers M rows, bit
+  // w>>2 selects which 16-wide K half it computes (acc[2][1]) => 2 waves per
+  // SIMD for ds_read/MFMA latency hiding at the same LDS footprint.
+  constexpr int NT = WAVES * 64;
+  constexpr int QS = (WAVES == 8) ? 1 : 2;  // q slots per wave
   int tile = blockIdx.x;
   int tile_w = W + 2;
   int tid = threadIdx.x;
   int wave = tid >> 6, lane = tid & 63;
+  int wave_m = (WAVES == 8) ? (wave & 3) : wave;
+  int qbase = (WAVES == 8) ? (wave >> 2) : 0;
   int n_cchunks = (C + CONV_CB - 1) / CONV_CB;
   int kz = K / CONVC_KB;
 
@@ -612,15 +621,15 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kzloop_kernel(
   // stage ALL input c-chunks once + the first weight slab
   for (int ci = 0; ci < n_cchunks; ++ci)
     issue_glds_input(x, s_convd + (size_t)ci * (CONVD_IN_CAP * 8), n0, h0, sb, bh, H, W, C,
-                     ci * CONV_CB, tile_w, tid, CONV_THREADS);
+                     ci * CONV_CB, tile_w, tid, NT);
   issue_glds_slab(wimg, s_w0, CONVC_WCHUNKS, tid);
   __syncthreads();
 
-  f32x4 acc[2][2];
+  f32x4 acc[2][QS];
 #pragma unroll
   for (int t = 0; t < 2; ++t)
 #pragma unroll
-    for (int q = 0; q < 2; ++q) acc[t][q] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int q = 0; q < QS; ++q) acc[t][q] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   int total = kz * n_cchunks;
   int cur = 0;
@@ -641,7 +650,7 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kzloop_kernel(
         bf16x8 afrag[2];
 #pragma unroll
         for (int t = 0; t < 2; ++t) {
-          int m = wave * 32 + t * 16 + (lane & 15);
+          int m = wave_m * 32 + t * 16 + (lane & 15);
           int sidx = m / pps, rem = m % pps;
           int hh = rem / W, ww = rem % W;
           int apos = (sidx * (bh + 2) + hh + dy) * tile_w + (ww + dx);
@@ -651,8 +660,8 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kzloop_kernel(
           afrag[t] = a;
         }
 #pragma unroll
-        for (int q = 0; q < 2; ++q) {
-          int br = tap * CONVC_KB + q * 16 + (lane & 15);
+        for (int q = 0; q < QS; ++q) {
+          int br = tap * CONVC_KB + (qbase + q) * 16 + (lane & 15);
           bf16x8 bfrag =
               *reinterpret_cast<const bf16x8*>(&sw[br * CONV_CB + (kbase ^ conv_swz(br))]);
 #pragma unroll
@@ -666,11 +675,11 @@ __global__ __launch_bounds__(CONV_THREADS) void conv3x3_fwd_kzloop_kernel(
 #pragma unroll
       for (int t = 0; t < 2; ++t)
 #pragma unroll
-        for (int q = 0; q < 2; ++q) {
+        for (int q = 0; q < QS; ++q) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            int m = wave * 32 + t * 16 + (lane >> 4) * 4 + r;
-            int k = q * 16 + (lane & 15);
+            int m = wave_m * 32 + t * 16 + (lane >> 4) * 4 + r;
+            int k = (qbase + q) * 16 + (lane & 15);
             if (m < m_count && kb0 + k < K) {
               int sidx = m / pps, rem = m % pps;
               int hh = rem / W, ww = rem % W;
@@ -693,7 +702,17 @@ extern "C" void launch_conv3x3_fwd_kzloop(const void* x, const void* wimg, const
   int BH, SB, h_groups, n_tiles;
   conv_tile_geom(Nn, H, W, 128, BH, SB, h_groups, n_tiles);
   dim3 grid(n_tiles, 1, 1);
-  conv3x3_fwd_kzloop_kernel<<<grid, CONV_THREADS, 0, s>>>(
-      (const bf16*)x, (const bf16*)wimg, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
-      h_groups);
+  static int waves = [] {
+    const char* e = getenv("FL4_KZLOOP_WAVES");
+    return e ? atoi(e) : 8;
+  }();
+  if (waves == 8) {
+    conv3x3_fwd_kzloop_kernel<8><<<grid, 512, 0, s>>>(
+        (const bf16*)x, (const bf16*)wimg, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
+        h_groups);
+  } else {
+    conv3x3_fwd_kzloop_kernel<4><<<grid, CONV_THREADS, 0, s>>>(
+        (const bf16*)x, (const bf16*)wimg, bias, (bf16*)y, Nn, H, W, C, K, BH, SB, n_tiles,
+        h_groups);
+  }
 }
