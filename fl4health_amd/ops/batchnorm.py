@@ -17,9 +17,11 @@ from fl4health_amd.ops import functional as F
 
 class _CdnaBatchNormFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu):
+    def forward(ctx, x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu, nbt=None):
         F._require_ext("bn_fwd_train")
-        y, mean, invstd = F._C.bn_fwd_train(x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu, None)
+        y, mean, invstd = F._C.bn_fwd_train(
+            x2d, gamma, beta, running_mean, running_var, momentum, eps, fuse_relu, None, nbt
+        )
         ctx.save_for_backward(x2d, mean, invstd, gamma, beta)
         ctx.fuse_relu = fuse_relu
         return y
@@ -28,7 +30,7 @@ class _CdnaBatchNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x2d, mean, invstd, gamma, beta = ctx.saved_tensors
         dx, dgamma, dbeta = F._C.bn_bwd(x2d, dy.contiguous(), mean, invstd, gamma, beta, ctx.fuse_relu, None)
-        return dx, dgamma, dbeta, None, None, None, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None, None
 
 
 class _CdnaBatchNormAddReluFn(torch.autograd.Function):
@@ -37,10 +39,10 @@ class _CdnaBatchNormAddReluFn(torch.autograd.Function):
     backward emits BOTH dx (through BN) and the residual's masked dy."""
 
     @staticmethod
-    def forward(ctx, x2d, res2d, gamma, beta, running_mean, running_var, momentum, eps):
+    def forward(ctx, x2d, res2d, gamma, beta, running_mean, running_var, momentum, eps, nbt=None):
         F._require_ext("bn_fwd_train")
         y, mean, invstd = F._C.bn_fwd_train(
-            x2d, gamma, beta, running_mean, running_var, momentum, eps, True, res2d
+            x2d, gamma, beta, running_mean, running_var, momentum, eps, True, res2d, nbt
         )
         ctx.save_for_backward(x2d, res2d, mean, invstd, gamma, beta)
         return y
@@ -51,7 +53,7 @@ class _CdnaBatchNormAddReluFn(torch.autograd.Function):
         dx, dgamma, dbeta, dres = F._C.bn_bwd(
             x2d, dy.contiguous(), mean, invstd, gamma, beta, True, res2d
         )
-        return dx, dres, dgamma, dbeta, None, None, None, None
+        return dx, dres, dgamma, dbeta, None, None, None, None, None
 
 
 class CdnaBatchNorm2d(nn.BatchNorm2d):
@@ -79,11 +81,10 @@ class CdnaBatchNorm2d(nn.BatchNorm2d):
         n, c, h, w = input.shape
         # channels-last memory IS [N*H*W, C] row-major
         x2d = input.permute(0, 2, 3, 1).reshape(n * h * w, c)
-        if self.num_batches_tracked is not None:
-            self.num_batches_tracked.add_(1)
+        # num_batches_tracked rides the finalize kernel (one fewer launch)
         y2d = _CdnaBatchNormFn.apply(
             x2d, self.weight.float(), self.bias.float(), self.running_mean, self.running_var,
-            float(self.momentum), float(self.eps), self.fuse_relu,
+            float(self.momentum), float(self.eps), self.fuse_relu, self.num_batches_tracked,
         )
         return y2d.view(n, h, w, c).permute(0, 3, 1, 2)
 
@@ -111,11 +112,9 @@ class CdnaBatchNorm2d(nn.BatchNorm2d):
             if residual.is_contiguous(memory_format=torch.channels_last)
             else residual.contiguous(memory_format=torch.channels_last).permute(0, 2, 3, 1).reshape(n * h * w, c)
         )
-        if self.num_batches_tracked is not None:
-            self.num_batches_tracked.add_(1)
         y2d = _CdnaBatchNormAddReluFn.apply(
             x2d, res2d, self.weight.float(), self.bias.float(), self.running_mean,
-            self.running_var, float(self.momentum), float(self.eps),
+            self.running_var, float(self.momentum), float(self.eps), self.num_batches_tracked,
         )
         return y2d.view(n, h, w, c).permute(0, 3, 1, 2)
 

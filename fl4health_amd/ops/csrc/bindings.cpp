@@ -27,7 +27,8 @@ void launch_confusion(const int64_t*, const int64_t*, unsigned long long*, int, 
                       hipStream_t);
 void launch_weighted_sum_rows(const float*, const float*, float*, int, int64_t, hipStream_t);
 void launch_bn_fwd(const void*, void*, float*, float*, float*, const float*, const float*, float*,
-                   float*, float, float, int64_t, int, int, int, int, const void*, hipStream_t);
+                   float*, int64_t*, float, float, int64_t, int, int, int, int, const void*,
+                   hipStream_t);
 void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int, int, int, int,
                              int, hipStream_t);
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
@@ -230,14 +231,17 @@ torch::Tensor weighted_sum_rows(torch::Tensor stack, torch::Tensor w) {
   return out;
 }
 
-// Row-group count: a pure function of the shape (deterministic), sized so the
-// grid covers all 256 CUs many times over (~4096 blocks regardless of C).
+// Partial-group (reduce BLOCK) count Gb: a pure function of the shape
+// (deterministic). Each reduce block is 128 channels x 8 row-stripes
+// (BNRW in bn_ops.hip), so thread occupancy matches the old 1-stripe
+// 4096-block grid while the partial array (and the finalize kernel's read
+// volume) shrinks 8x.
 int bn_groups(int64_t R, int C) {
   int cblocks = (C + 127) / 128;
-  int64_t g = 4096 / cblocks;
-  if (g > (R + 3) / 4) g = (R + 3) / 4;  // at least ~4 rows per group
+  int64_t g = 512 / cblocks;
+  if (g > (R + 31) / 32) g = (R + 31) / 32;  // >= ~4 rows per stripe
   if (g < 1) g = 1;
-  if (g > 8192) g = 8192;
+  if (g > 1024) g = 1024;
   return (int)g;
 }
 
@@ -253,7 +257,8 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
                                         c10::optional<torch::Tensor> running_mean,
                                         c10::optional<torch::Tensor> running_var, double momentum,
                                         double eps, bool fuse_relu,
-                                        c10::optional<torch::Tensor> res) {
+                                        c10::optional<torch::Tensor> res,
+                                        c10::optional<torch::Tensor> num_batches_tracked) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2, "x must be [R, C] contiguous");
   int64_t R = x.size(0);
   int C = (int)x.size(1);
@@ -270,8 +275,9 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Te
       gamma.has_value() ? gamma->data_ptr<float>() : nullptr,
       beta.has_value() ? beta->data_ptr<float>() : nullptr,
       running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
-      running_var.has_value() ? running_var->data_ptr<float>() : nullptr, (float)momentum,
-      (float)eps, R, C, G, dtype, fuse_relu ? 1 : 0,
+      running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
+      num_batches_tracked.has_value() ? num_batches_tracked->data_ptr<int64_t>() : nullptr,
+      (float)momentum, (float)eps, R, C, G, dtype, fuse_relu ? 1 : 0,
       res.has_value() ? res->data_ptr() : nullptr, stream());
   return {y, mean, invstd};
 }
@@ -523,7 +529,10 @@ torch::Tensor mkmmd_backward(torch::Tensor d, torch::Tensor gammas, torch::Tenso
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("bn_fwd_train", &bn_fwd_train, "NHWC batchnorm training forward");
+  m.def("bn_fwd_train", &bn_fwd_train, "NHWC batchnorm training forward", py::arg("x"),
+        py::arg("gamma"), py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
+        py::arg("momentum"), py::arg("eps"), py::arg("fuse_relu"), py::arg("res"),
+        py::arg("num_batches_tracked") = py::none());
   m.def("bn_bwd", &bn_bwd, "NHWC batchnorm backward");
   m.def("axpby_", &axpby_, "y = a*x + b*y (in-place)");
   m.def("prox_sgd_step_", &prox_sgd_step_, "fused proximal SGD step");

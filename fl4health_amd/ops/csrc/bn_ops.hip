@@ -20,6 +20,10 @@
 #include <algorithm>
 
 #define BNBLOCK 128
+// Row-stripes per reduce block: each block LDS-combines BNRW per-thread
+// stripe sums before writing ONE partial row, shrinking the partial array
+// (and the finalize kernel's read volume) by BNRW vs one-stripe blocks.
+#define BNRW 8
 
 template <typename T>
 __device__ __forceinline__ float ld(const T* p);
@@ -40,23 +44,37 @@ __device__ __forceinline__ void st<__hip_bfloat16>(__hip_bfloat16* p, float v) {
 }
 
 // ---------------------------------------------------------------------------
-// fwd pass 1: per-channel partial sums over fixed G row-groups
-// partial layout: [2, G, C] (sum, sumsq)
+// fwd pass 1: per-channel partial sums. Deterministic fixed-stripe split:
+// thread (by, ty) owns rows r == by*BNRW+ty (mod Gb*BNRW); the block combines
+// its BNRW stripes through LDS and writes one row of partial [2, Gb, C].
 // ---------------------------------------------------------------------------
 template <typename T>
-__global__ __launch_bounds__(BNBLOCK) void bn_fwd_reduce_kernel(
-    const T* __restrict__ x, float* __restrict__ partial, int64_t R, int C, int G) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  int g = blockIdx.y;
+__global__ __launch_bounds__(BNBLOCK * BNRW) void bn_fwd_reduce_kernel(
+    const T* __restrict__ x, float* __restrict__ partial, int64_t R, int C, int Gb) {
+  __shared__ float sm[BNRW][2][BNBLOCK];
+  int c = blockIdx.x * BNBLOCK + threadIdx.x;
+  int S = Gb * BNRW;
+  int stripe = blockIdx.y * BNRW + threadIdx.y;
   float s = 0.0f, ss = 0.0f;
-  for (int64_t r = g; r < R; r += G) {
-    float v = ld<T>(x + r * C + c);
-    s += v;
-    ss = fmaf(v, v, ss);
+  if (c < C) {
+    for (int64_t r = stripe; r < R; r += S) {
+      float v = ld<T>(x + r * C + c);
+      s += v;
+      ss = fmaf(v, v, ss);
+    }
   }
-  partial[(int64_t)g * C + c] = s;
-  partial[(int64_t)(G + g) * C + c] = ss;
+  sm[threadIdx.y][0][threadIdx.x] = s;
+  sm[threadIdx.y][1][threadIdx.x] = ss;
+  __syncthreads();
+  if (threadIdx.y == 0 && c < C) {
+#pragma unroll
+    for (int j = 1; j < BNRW; ++j) {
+      s += sm[j][0][threadIdx.x];
+      ss += sm[j][1][threadIdx.x];
+    }
+    partial[(int64_t)blockIdx.y * C + c] = s;
+    partial[(int64_t)(Gb + blockIdx.y) * C + c] = ss;
+  }
 }
 
 // fwd pass 2: finalize mean/invstd, update running stats.
@@ -65,8 +83,13 @@ __global__ __launch_bounds__(BNBLOCK) void bn_fwd_reduce_kernel(
 #define BNFIN 256
 __global__ __launch_bounds__(BNFIN) void bn_fwd_finalize_kernel(
     const float* __restrict__ partial, float* __restrict__ mean, float* __restrict__ invstd,
-    float* __restrict__ running_mean, float* __restrict__ running_var, float momentum, float eps,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    int64_t* __restrict__ num_batches_tracked, float momentum, float eps,
     int64_t R, int C, int G) {
+  // buffer-update fold: the counter bump that was a separate 4 us
+  // CUDAFunctorOnSelf_add<long> launch per BN layer per step
+  if (blockIdx.x == 0 && threadIdx.x == 0 && num_batches_tracked != nullptr)
+    ++*num_batches_tracked;
   __shared__ float sm_s[BNFIN];
   __shared__ float sm_ss[BNFIN];
   int c = blockIdx.x;
@@ -125,34 +148,47 @@ __global__ __launch_bounds__(BNBLOCK) void bn_fwd_norm_kernel(
 // bwd pass 1: per-channel partials of (sum dy, sum dy * x_hat)
 // ---------------------------------------------------------------------------
 template <typename T>
-__global__ __launch_bounds__(BNBLOCK) void bn_bwd_reduce_kernel(
+__global__ __launch_bounds__(BNBLOCK * BNRW) void bn_bwd_reduce_kernel(
     const T* __restrict__ x, const T* __restrict__ dy, float* __restrict__ partial,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta, const T* __restrict__ res,
-    T* __restrict__ dres, int fuse_relu, int64_t R, int C, int G) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  int g = blockIdx.y;
-  float m = mean[c];
-  float is = invstd[c];
-  // with fused ReLU, dy must be masked where the pre-activation was <= 0:
-  // recompute the sign from (x_hat * gamma + beta) instead of saving y
-  float gm = gamma != nullptr ? gamma[c] : 1.0f;
-  float bt = beta != nullptr ? beta[c] : 0.0f;
+    T* __restrict__ dres, int fuse_relu, int64_t R, int C, int Gb) {
+  __shared__ float sm[BNRW][2][BNBLOCK];
+  int c = blockIdx.x * BNBLOCK + threadIdx.x;
+  int S = Gb * BNRW;
+  int stripe = blockIdx.y * BNRW + threadIdx.y;
   float sdy = 0.0f, sdyx = 0.0f;
-  for (int64_t r = g; r < R; r += G) {
-    float gy = ld<T>(dy + r * C + c);
-    float xh = (ld<T>(x + r * C + c) - m) * is;
-    float pre = fmaf(xh, gm, bt);
-    if (res != nullptr) pre += ld<T>(res + r * C + c);
-    if (fuse_relu && pre <= 0.0f) gy = 0.0f;
-    // the residual branch receives exactly the relu-masked dy
-    if (dres != nullptr) st<T>(dres + r * C + c, gy);
-    sdy += gy;
-    sdyx = fmaf(gy, xh, sdyx);
+  if (c < C) {
+    float m = mean[c];
+    float is = invstd[c];
+    // with fused ReLU, dy must be masked where the pre-activation was <= 0:
+    // recompute the sign from (x_hat * gamma + beta) instead of saving y
+    float gm = gamma != nullptr ? gamma[c] : 1.0f;
+    float bt = beta != nullptr ? beta[c] : 0.0f;
+    for (int64_t r = stripe; r < R; r += S) {
+      float gy = ld<T>(dy + r * C + c);
+      float xh = (ld<T>(x + r * C + c) - m) * is;
+      float pre = fmaf(xh, gm, bt);
+      if (res != nullptr) pre += ld<T>(res + r * C + c);
+      if (fuse_relu && pre <= 0.0f) gy = 0.0f;
+      // the residual branch receives exactly the relu-masked dy
+      if (dres != nullptr) st<T>(dres + r * C + c, gy);
+      sdy += gy;
+      sdyx = fmaf(gy, xh, sdyx);
+    }
   }
-  partial[(int64_t)g * C + c] = sdy;
-  partial[(int64_t)(G + g) * C + c] = sdyx;
+  sm[threadIdx.y][0][threadIdx.x] = sdy;
+  sm[threadIdx.y][1][threadIdx.x] = sdyx;
+  __syncthreads();
+  if (threadIdx.y == 0 && c < C) {
+#pragma unroll
+    for (int j = 1; j < BNRW; ++j) {
+      sdy += sm[j][0][threadIdx.x];
+      sdyx += sm[j][1][threadIdx.x];
+    }
+    partial[(int64_t)blockIdx.y * C + c] = sdy;
+    partial[(int64_t)(Gb + blockIdx.y) * C + c] = sdyx;
+  }
 }
 
 __global__ __launch_bounds__(BNFIN) void bn_bwd_finalize_kernel(
@@ -224,21 +260,35 @@ static inline void bn_dims(int C, int64_t R, int G, dim3* grid, dim3* block) {
   grid->z = 1;
 }
 
+// G here is the partial-group count Gb; the reduce kernels internally stripe
+// rows S = Gb*BNRW ways and LDS-combine, so thread count matches the 1D form.
+static inline void bn_reduce_dims(int C, int G, dim3* grid, dim3* block) {
+  block->x = BNBLOCK;
+  block->y = BNRW;
+  block->z = 1;
+  grid->x = (C + BNBLOCK - 1) / BNBLOCK;
+  grid->y = G;
+  grid->z = 1;
+}
+
 extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mean, float* invstd,
                               const float* gamma, const float* beta, float* running_mean,
-                              float* running_var, float momentum, float eps, int64_t R, int C,
+                              float* running_var, int64_t* num_batches_tracked, float momentum,
+                              float eps, int64_t R, int C,
                               int G, int dtype, int fuse_relu, const void* res, hipStream_t s) {
-  dim3 grid, block;
-  bn_dims(C, R, G, &grid, &block);
+  dim3 grid, block, rgrid, rblock;
+  bn_dims(C, R, G * BNRW, &grid, &block);
+  bn_reduce_dims(C, G, &rgrid, &rblock);
   if (dtype == 1) {
-    bn_fwd_reduce_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
+    bn_fwd_reduce_kernel<__hip_bfloat16><<<rgrid, rblock, 0, s>>>(
         (const __hip_bfloat16*)x, partial, R, C, G);
   } else {
-    bn_fwd_reduce_kernel<float><<<grid, block, 0, s>>>((const float*)x, partial, R, C, G);
+    bn_fwd_reduce_kernel<float><<<rgrid, rblock, 0, s>>>((const float*)x, partial, R, C, G);
   }
   bn_fwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
-      partial, mean, invstd, running_mean, running_var, momentum, eps, R, C, G);
-  // normalize: reuse G-deep row grid (bandwidth-bound)
+      partial, mean, invstd, running_mean, running_var, num_batches_tracked, momentum, eps, R, C,
+      G);
+  // normalize: G*BNRW-deep row grid (bandwidth-bound)
   if (dtype == 1) {
     bn_fwd_norm_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, invstd, gamma, beta,
@@ -255,14 +305,15 @@ extern "C" void launch_bn_bwd(const void* x, const void* dy, void* dx, float* pa
                               const float* beta, float* sum_dy, float* sum_dy_xhat, float* dgamma,
                               float* dbeta, int64_t R, int C, int G, int dtype, int fuse_relu,
                               const void* res, void* dres, hipStream_t s) {
-  dim3 grid, block;
-  bn_dims(C, R, G, &grid, &block);
+  dim3 grid, block, rgrid, rblock;
+  bn_dims(C, R, G * BNRW, &grid, &block);
+  bn_reduce_dims(C, G, &rgrid, &rblock);
   if (dtype == 1) {
-    bn_bwd_reduce_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
+    bn_bwd_reduce_kernel<__hip_bfloat16><<<rgrid, rblock, 0, s>>>(
         (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, partial, mean, invstd, gamma, beta,
         (const __hip_bfloat16*)res, (__hip_bfloat16*)dres, fuse_relu, R, C, G);
   } else {
-    bn_bwd_reduce_kernel<float><<<grid, block, 0, s>>>((const float*)x, (const float*)dy, partial,
+    bn_bwd_reduce_kernel<float><<<rgrid, rblock, 0, s>>>((const float*)x, (const float*)dy, partial,
                                                        mean, invstd, gamma, beta,
                                                        (const float*)res, (float*)dres, fuse_relu,
                                                        R, C, G);
