@@ -310,6 +310,12 @@ class BasicClient:
     def train_step(self, input: TorchInputType, target: TorchTargetType) -> tuple[TrainingLosses, TorchPredType]:
         """forward -> loss -> backward -> transform_gradients -> step (reference :578-603)."""
         self.set_optimizer_zero_grad()
+        losses, preds = self._forward_backward(input, target)
+        self.transform_gradients(losses)
+        self.step_optimizers()
+        return losses, preds
+
+    def _forward_backward(self, input: TorchInputType, target: TorchTargetType) -> tuple[TrainingLosses, TorchPredType]:
         if self.autocast_dtype is not None and self.device.type == "cuda":
             with torch.autocast(device_type="cuda", dtype=self.autocast_dtype):
                 preds, features = self.predict(input)
@@ -320,8 +326,6 @@ class BasicClient:
             target = self.transform_target(target)
             losses = self.compute_training_loss(preds, features, target)
         losses.backward["backward"].backward()
-        self.transform_gradients(losses)
-        self.step_optimizers()
         return losses, preds
 
     def set_optimizer_zero_grad(self) -> None:
@@ -352,6 +356,42 @@ class BasicClient:
         self._graph.replay()
         return st["losses"], st["preds"]
 
+    @staticmethod
+    def _grad_as_1d(t: torch.Tensor) -> torch.Tensor:
+        """Contiguous 1-D alias of a grad tensor. Flat-bound NHWC conv views
+        (NCHW-logical permutes of a flat slice) and channels-last backward
+        outputs share the same memory order, so both sides reduce to plain
+        contiguous vectors and the batched copy stays on the foreach fast
+        path."""
+        if t.dim() == 4 and not t.is_contiguous() and t.is_contiguous(memory_format=torch.channels_last):
+            return t.permute(0, 2, 3, 1).reshape(-1)
+        return t.reshape(-1)
+
+    def _graph_train_step(self, input: torch.Tensor, target: torch.Tensor):
+        """train_step body used ONLY under hipGraph capture: steal-then-pack
+        gradient flow. With flat-bound grads, eager steps pay zero_grad fill +
+        AccumulateGrad add per parameter (2 elementwise kernels each, ~73
+        small launches/step on ResNet-18). Inside a graph the backward's
+        output buffers live at stable pool addresses, so we let AccumulateGrad
+        steal them (p.grad=None first: no fill, no add) and pack all of them
+        into the flat grad buffers with one batched foreach copy before the
+        fused optimizer step."""
+        views = [(p, p.grad) for p in self.model.parameters() if p.requires_grad and p.grad is not None]
+        for p, _ in views:
+            p.grad = None
+        losses, preds = self._forward_backward(input, target)
+        srcs, dsts = [], []
+        for p, v in views:
+            g = p.grad if p.grad is not None else torch.zeros_like(v)
+            srcs.append(self._grad_as_1d(g))
+            dsts.append(self._grad_as_1d(v))
+            p.grad = v  # restore flat aliasing for transform/step/exchange
+        if srcs:
+            torch._foreach_copy_(dsts, srcs)
+        self.transform_gradients(losses)
+        self.step_optimizers()
+        return losses, preds, srcs
+
     def _capture_train_graph(self, input: torch.Tensor, target: torch.Tensor) -> None:
         st: dict[str, Any] = {"input": input.clone(), "target": target.clone()}
         side = torch.cuda.Stream()
@@ -360,9 +400,16 @@ class BasicClient:
             for _ in range(3):  # warmup: MIOpen algo find + allocator steady-state
                 self.train_step(st["input"], st["target"])
         torch.cuda.current_stream().wait_stream(side)
+        # Subclasses with their own train_step keep their logic under capture;
+        # only the stock body opts into the steal-then-pack grad flow.
+        stock = type(self).train_step is BasicClient.train_step
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            losses, preds = self.train_step(st["input"], st["target"])
+            if stock:
+                losses, preds, stolen = self._graph_train_step(st["input"], st["target"])
+                st["stolen"] = stolen  # keep pool buffers alive for replays
+            else:
+                losses, preds = self.train_step(st["input"], st["target"])
         st["losses"], st["preds"] = losses, preds
         self._graph, self._graph_static = graph, st
         log.info("Client %s: captured hipGraph train step (batch %s)", self.client_name, tuple(input.shape))
