@@ -386,8 +386,14 @@ class BasicClient:
             srcs.append(self._grad_as_1d(g))
             dsts.append(self._grad_as_1d(v))
             p.grad = v  # restore flat aliasing for transform/step/exchange
-        if srcs:
-            torch._foreach_copy_(dsts, srcs)
+        # one foreach per dtype: a mixed-dtype list knocks _foreach_copy_ off
+        # its fused multi-tensor kernel into per-pair hipMemcpy
+        by_dtype: dict[torch.dtype, tuple[list, list]] = {}
+        for d, s in zip(dsts, srcs):
+            by_dtype.setdefault(d.dtype, ([], []))[0].append(d)
+            by_dtype.setdefault(d.dtype, ([], []))[1].append(s.to(d.dtype) if s.dtype != d.dtype else s)
+        for dl, sl in by_dtype.values():
+            torch._foreach_copy_(dl, sl)
         self.transform_gradients(losses)
         self.step_optimizers()
         return losses, preds, srcs
