@@ -248,6 +248,270 @@ __global__ __launch_bounds__(BNBLOCK) void bn_bwd_dx_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// v2 kernels: channel-PAIR layout. Block (PX pair-slots, PY row-stripes) with
+// PX*PY = 1024; every thread owns 2 adjacent channels (one 4B/8B vector load
+// per row visit) and ALL threads stay active even at C=64, where the 1D
+// 128-channel block wasted half its lanes on 2-byte scalar loads.
+// ---------------------------------------------------------------------------
+struct P2 {
+  float a, b;
+};
+
+template <typename T>
+__device__ __forceinline__ P2 ldp(const T* p);
+template <>
+__device__ __forceinline__ P2 ldp<float>(const float* p) {
+  float2 v = *reinterpret_cast<const float2*>(p);
+  return P2{v.x, v.y};
+}
+template <>
+__device__ __forceinline__ P2 ldp<__hip_bfloat16>(const __hip_bfloat16* p) {
+  uint32_t u = *reinterpret_cast<const uint32_t*>(p);
+  union {
+    uint32_t u;
+    __hip_bfloat16 h[2];
+  } c{u};
+  return P2{__bfloat162float(c.h[0]), __bfloat162float(c.h[1])};
+}
+
+template <typename T>
+__device__ __forceinline__ void stp(T* p, float a, float b);
+template <>
+__device__ __forceinline__ void stp<float>(float* p, float a, float b) {
+  *reinterpret_cast<float2*>(p) = float2{a, b};
+}
+template <>
+__device__ __forceinline__ void stp<__hip_bfloat16>(__hip_bfloat16* p, float a, float b) {
+  union {
+    uint32_t u;
+    __hip_bfloat16 h[2];
+  } c;
+  c.h[0] = __float2bfloat16(a);
+  c.h[1] = __float2bfloat16(b);
+  *reinterpret_cast<uint32_t*>(p) = c.u;
+}
+
+// pure geometry shared with the binding (keep in sync with bindings.cpp):
+// PX = pow2 >= C/2 capped at 128; PY = 1024 / PX
+static __host__ __device__ inline void bn_geom(int C, int* px, int* py) {
+  int half = C >> 1;
+  int p = 8;
+  while (p < half && p < 128) p <<= 1;
+  *px = p;
+  *py = 1024 / p;
+}
+
+template <typename T>
+__global__ __launch_bounds__(1024) void bn_fwd_reduce2_kernel(
+    const T* __restrict__ x, float* __restrict__ partial, int64_t R, int C, int Gb) {
+  __shared__ float sm[2][1024];
+  int half = C >> 1;
+  int cp = blockIdx.x * blockDim.x + threadIdx.x;
+  int S = Gb * blockDim.y;
+  int stripe = blockIdx.y * blockDim.y + threadIdx.y;
+  float s0 = 0.f, ss0 = 0.f, s1 = 0.f, ss1 = 0.f;
+  if (cp < half) {
+    const T* base = x + 2 * cp;
+    for (int64_t r = stripe; r < R; r += S) {
+      P2 v = ldp<T>(base + r * C);
+      s0 += v.a;
+      ss0 = fmaf(v.a, v.a, ss0);
+      s1 += v.b;
+      ss1 = fmaf(v.b, v.b, ss1);
+    }
+  }
+  int tid = threadIdx.y * blockDim.x + threadIdx.x;
+  // two LDS rounds (sum then sumsq pairs) keep LDS at 8 KB
+  sm[0][tid] = s0;
+  sm[1][tid] = s1;
+  __syncthreads();
+  for (int off = blockDim.y >> 1; off > 0; off >>= 1) {
+    if (threadIdx.y < off) {
+      sm[0][tid] += sm[0][tid + off * blockDim.x];
+      sm[1][tid] += sm[1][tid + off * blockDim.x];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.y == 0 && cp < half) {
+    partial[(int64_t)blockIdx.y * C + 2 * cp] = sm[0][threadIdx.x];
+    partial[(int64_t)blockIdx.y * C + 2 * cp + 1] = sm[1][threadIdx.x];
+  }
+  __syncthreads();
+  sm[0][tid] = ss0;
+  sm[1][tid] = ss1;
+  __syncthreads();
+  for (int off = blockDim.y >> 1; off > 0; off >>= 1) {
+    if (threadIdx.y < off) {
+      sm[0][tid] += sm[0][tid + off * blockDim.x];
+      sm[1][tid] += sm[1][tid + off * blockDim.x];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.y == 0 && cp < half) {
+    partial[(int64_t)(Gb + blockIdx.y) * C + 2 * cp] = sm[0][threadIdx.x];
+    partial[(int64_t)(Gb + blockIdx.y) * C + 2 * cp + 1] = sm[1][threadIdx.x];
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(1024) void bn_fwd_norm2_kernel(
+    const T* __restrict__ x, T* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, const T* __restrict__ res, int fuse_relu, int64_t R, int C) {
+  int half = C >> 1;
+  int cp = blockIdx.x * blockDim.x + threadIdx.x;
+  if (cp >= half) return;
+  int c0 = 2 * cp;
+  float is0 = invstd[c0], is1 = invstd[c0 + 1];
+  float g0 = gamma != nullptr ? gamma[c0] : 1.0f;
+  float g1 = gamma != nullptr ? gamma[c0 + 1] : 1.0f;
+  float b0 = beta != nullptr ? beta[c0] : 0.0f;
+  float b1 = beta != nullptr ? beta[c0 + 1] : 0.0f;
+  float sc0 = is0 * g0, sc1 = is1 * g1;
+  float sh0 = b0 - mean[c0] * sc0, sh1 = b1 - mean[c0 + 1] * sc1;
+  int S = gridDim.y * blockDim.y;
+  const T* xb = x + c0;
+  const T* rb = res != nullptr ? res + c0 : nullptr;
+  T* yb = y + c0;
+  for (int64_t r = blockIdx.y * blockDim.y + threadIdx.y; r < R; r += S) {
+    P2 v = ldp<T>(xb + r * C);
+    float o0 = fmaf(v.a, sc0, sh0);
+    float o1 = fmaf(v.b, sc1, sh1);
+    if (rb != nullptr) {
+      P2 rv = ldp<T>(rb + r * C);
+      o0 += rv.a;
+      o1 += rv.b;
+    }
+    if (fuse_relu) {
+      o0 = fmaxf(o0, 0.0f);
+      o1 = fmaxf(o1, 0.0f);
+    }
+    stp<T>(yb + r * C, o0, o1);
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(1024) void bn_bwd_reduce2_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, float* __restrict__ partial,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta, const T* __restrict__ res,
+    T* __restrict__ dres, int fuse_relu, int64_t R, int C, int Gb) {
+  __shared__ float sm[2][1024];
+  int half = C >> 1;
+  int cp = blockIdx.x * blockDim.x + threadIdx.x;
+  int S = Gb * blockDim.y;
+  int stripe = blockIdx.y * blockDim.y + threadIdx.y;
+  float sdy0 = 0.f, sdyx0 = 0.f, sdy1 = 0.f, sdyx1 = 0.f;
+  if (cp < half) {
+    int c0 = 2 * cp;
+    float m0 = mean[c0], m1 = mean[c0 + 1];
+    float is0 = invstd[c0], is1 = invstd[c0 + 1];
+    float g0 = gamma != nullptr ? gamma[c0] : 1.0f;
+    float g1 = gamma != nullptr ? gamma[c0 + 1] : 1.0f;
+    float b0 = beta != nullptr ? beta[c0] : 0.0f;
+    float b1 = beta != nullptr ? beta[c0 + 1] : 0.0f;
+    const T* xb = x + c0;
+    const T* dyb = dy + c0;
+    const T* rb = res != nullptr ? res + c0 : nullptr;
+    T* drb = dres != nullptr ? dres + c0 : nullptr;
+    for (int64_t r = stripe; r < R; r += S) {
+      P2 gy = ldp<T>(dyb + r * C);
+      P2 xv = ldp<T>(xb + r * C);
+      float xh0 = (xv.a - m0) * is0, xh1 = (xv.b - m1) * is1;
+      float pre0 = fmaf(xh0, g0, b0), pre1 = fmaf(xh1, g1, b1);
+      if (rb != nullptr) {
+        P2 rv = ldp<T>(rb + r * C);
+        pre0 += rv.a;
+        pre1 += rv.b;
+      }
+      if (fuse_relu) {
+        if (pre0 <= 0.0f) gy.a = 0.0f;
+        if (pre1 <= 0.0f) gy.b = 0.0f;
+      }
+      if (drb != nullptr) stp<T>(drb + r * C, gy.a, gy.b);
+      sdy0 += gy.a;
+      sdyx0 = fmaf(gy.a, xh0, sdyx0);
+      sdy1 += gy.b;
+      sdyx1 = fmaf(gy.b, xh1, sdyx1);
+    }
+  }
+  int tid = threadIdx.y * blockDim.x + threadIdx.x;
+  sm[0][tid] = sdy0;
+  sm[1][tid] = sdy1;
+  __syncthreads();
+  for (int off = blockDim.y >> 1; off > 0; off >>= 1) {
+    if (threadIdx.y < off) {
+      sm[0][tid] += sm[0][tid + off * blockDim.x];
+      sm[1][tid] += sm[1][tid + off * blockDim.x];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.y == 0 && cp < half) {
+    partial[(int64_t)blockIdx.y * C + 2 * cp] = sm[0][threadIdx.x];
+    partial[(int64_t)blockIdx.y * C + 2 * cp + 1] = sm[1][threadIdx.x];
+  }
+  __syncthreads();
+  sm[0][tid] = sdyx0;
+  sm[1][tid] = sdyx1;
+  __syncthreads();
+  for (int off = blockDim.y >> 1; off > 0; off >>= 1) {
+    if (threadIdx.y < off) {
+      sm[0][tid] += sm[0][tid + off * blockDim.x];
+      sm[1][tid] += sm[1][tid + off * blockDim.x];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.y == 0 && cp < half) {
+    partial[(int64_t)(Gb + blockIdx.y) * C + 2 * cp] = sm[0][threadIdx.x];
+    partial[(int64_t)(Gb + blockIdx.y) * C + 2 * cp + 1] = sm[1][threadIdx.x];
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(1024) void bn_bwd_dx2_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ sum_dy, const float* __restrict__ sum_dy_xhat,
+    const T* __restrict__ res, int fuse_relu, int64_t R, int C) {
+  int half = C >> 1;
+  int cp = blockIdx.x * blockDim.x + threadIdx.x;
+  if (cp >= half) return;
+  int c0 = 2 * cp;
+  float m0 = mean[c0], m1 = mean[c0 + 1];
+  float is0 = invstd[c0], is1 = invstd[c0 + 1];
+  float g0 = gamma != nullptr ? gamma[c0] : 1.0f;
+  float g1 = gamma != nullptr ? gamma[c0 + 1] : 1.0f;
+  float b0 = beta != nullptr ? beta[c0] : 0.0f;
+  float b1 = beta != nullptr ? beta[c0 + 1] : 0.0f;
+  float k0 = g0 * is0, k1 = g1 * is1;
+  float mdy0 = sum_dy[c0] / (float)R, mdy1 = sum_dy[c0 + 1] / (float)R;
+  float mdx0 = sum_dy_xhat[c0] / (float)R, mdx1 = sum_dy_xhat[c0 + 1] / (float)R;
+  int S = gridDim.y * blockDim.y;
+  const T* xb = x + c0;
+  const T* dyb = dy + c0;
+  const T* rb = res != nullptr ? res + c0 : nullptr;
+  T* dxb = dx + c0;
+  for (int64_t r = blockIdx.y * blockDim.y + threadIdx.y; r < R; r += S) {
+    P2 gy = ldp<T>(dyb + r * C);
+    P2 xv = ldp<T>(xb + r * C);
+    float xh0 = (xv.a - m0) * is0, xh1 = (xv.b - m1) * is1;
+    float pre0 = fmaf(xh0, g0, b0), pre1 = fmaf(xh1, g1, b1);
+    if (rb != nullptr) {
+      P2 rv = ldp<T>(rb + r * C);
+      pre0 += rv.a;
+      pre1 += rv.b;
+    }
+    if (fuse_relu) {
+      if (pre0 <= 0.0f) gy.a = 0.0f;
+      if (pre1 <= 0.0f) gy.b = 0.0f;
+    }
+    stp<T>(dxb + r * C, k0 * (gy.a - mdy0 - xh0 * mdx0), k1 * (gy.b - mdy1 - xh1 * mdx1));
+  }
+}
+
 // ---------------------------------------------------------------------------
 // launchers (dtype: 0 = fp32, 1 = bf16); G fixed for determinism
 // ---------------------------------------------------------------------------
@@ -276,6 +540,31 @@ extern "C" void launch_bn_fwd(const void* x, void* y, float* partial, float* mea
                               float* running_var, int64_t* num_batches_tracked, float momentum,
                               float eps, int64_t R, int C,
                               int G, int dtype, int fuse_relu, const void* res, hipStream_t s) {
+  if (C % 2 == 0) {
+    int px, py;
+    bn_geom(C, &px, &py);
+    int half = C >> 1;
+    dim3 rb2(px, py, 1), rg2((half + px - 1) / px, G, 1);
+    if (dtype == 1) {
+      bn_fwd_reduce2_kernel<__hip_bfloat16><<<rg2, rb2, 0, s>>>(
+          (const __hip_bfloat16*)x, partial, R, C, G);
+    } else {
+      bn_fwd_reduce2_kernel<float><<<rg2, rb2, 0, s>>>((const float*)x, partial, R, C, G);
+    }
+    bn_fwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
+        partial, mean, invstd, running_mean, running_var, num_batches_tracked, momentum, eps, R,
+        C, G);
+    if (dtype == 1) {
+      bn_fwd_norm2_kernel<__hip_bfloat16><<<rg2, rb2, 0, s>>>(
+          (const __hip_bfloat16*)x, (__hip_bfloat16*)y, mean, invstd, gamma, beta,
+          (const __hip_bfloat16*)res, fuse_relu, R, C);
+    } else {
+      bn_fwd_norm2_kernel<float><<<rg2, rb2, 0, s>>>((const float*)x, (float*)y, mean, invstd,
+                                                     gamma, beta, (const float*)res, fuse_relu,
+                                                     R, C);
+    }
+    return;
+  }
   dim3 grid, block, rgrid, rblock;
   bn_dims(C, R, G * BNRW, &grid, &block);
   bn_reduce_dims(C, G, &rgrid, &rblock);
@@ -305,6 +594,33 @@ extern "C" void launch_bn_bwd(const void* x, const void* dy, void* dx, float* pa
                               const float* beta, float* sum_dy, float* sum_dy_xhat, float* dgamma,
                               float* dbeta, int64_t R, int C, int G, int dtype, int fuse_relu,
                               const void* res, void* dres, hipStream_t s) {
+  if (C % 2 == 0) {
+    int px, py;
+    bn_geom(C, &px, &py);
+    int half = C >> 1;
+    dim3 rb2(px, py, 1), rg2((half + px - 1) / px, G, 1);
+    if (dtype == 1) {
+      bn_bwd_reduce2_kernel<__hip_bfloat16><<<rg2, rb2, 0, s>>>(
+          (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, partial, mean, invstd, gamma,
+          beta, (const __hip_bfloat16*)res, (__hip_bfloat16*)dres, fuse_relu, R, C, G);
+    } else {
+      bn_bwd_reduce2_kernel<float><<<rg2, rb2, 0, s>>>(
+          (const float*)x, (const float*)dy, partial, mean, invstd, gamma, beta,
+          (const float*)res, (float*)dres, fuse_relu, R, C, G);
+    }
+    bn_bwd_finalize_kernel<<<dim3(C, 1, 1), dim3(BNFIN, 1, 1), 0, s>>>(
+        partial, sum_dy, sum_dy_xhat, dgamma, dbeta, C, G);
+    if (dtype == 1) {
+      bn_bwd_dx2_kernel<__hip_bfloat16><<<rg2, rb2, 0, s>>>(
+          (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, (__hip_bfloat16*)dx, mean, invstd,
+          gamma, beta, sum_dy, sum_dy_xhat, (const __hip_bfloat16*)res, fuse_relu, R, C);
+    } else {
+      bn_bwd_dx2_kernel<float><<<rg2, rb2, 0, s>>>(
+          (const float*)x, (const float*)dy, (float*)dx, mean, invstd, gamma, beta, sum_dy,
+          sum_dy_xhat, (const float*)res, fuse_relu, R, C);
+    }
+    return;
+  }
   dim3 grid, block, rgrid, rblock;
   bn_dims(C, R, G * BNRW, &grid, &block);
   bn_reduce_dims(C, G, &rgrid, &rblock);
