@@ -1,5 +1,6 @@
 // Python bindings for the CDNA4 flat-buffer FL kernels (see flat_ops.hip).
 #include <torch/extension.h>
+#include <cstdlib>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
@@ -237,8 +238,12 @@ torch::Tensor weighted_sum_rows(torch::Tensor stack, torch::Tensor w) {
 // 4096-block grid while the partial array (and the finalize kernel's read
 // volume) shrinks 8x.
 int bn_groups(int64_t R, int C) {
+  static int target = [] {
+    const char* e = getenv("FL4_BN_GB");
+    return e ? atoi(e) : 512;
+  }();
   int cblocks = (C + 127) / 128;
-  int64_t g = 512 / cblocks;
+  int64_t g = target / cblocks;
   if (g > (R + 31) / 32) g = (R + 31) / 32;  // >= ~4 rows per stripe
   if (g < 1) g = 1;
   if (g > 1024) g = 1024;
