@@ -235,3 +235,23 @@ def test_dp_sgd_no_clip_no_noise_equals_sgd_cpu():
     eng.step()
     for p, pr in zip(model.parameters(), ref_model.parameters()):
         assert torch.allclose(p, pr, atol=1e-5), f"max diff {(p - pr).abs().max()}"
+
+
+def test_every_module_imports():
+    """Walk the whole package: every module must import (guards against
+    bitrot in less-exercised corners; gpu/optional deps are import-guarded
+    inside the modules themselves)."""
+    import importlib
+    import pkgutil
+
+    import fl4health_amd
+
+    failures = []
+    for info in pkgutil.walk_packages(fl4health_amd.__path__, prefix="fl4health_amd."):
+        if "csrc" in info.name:
+            continue
+        try:
+            importlib.import_module(info.name)
+        except Exception as e:  # noqa: BLE001 - collecting all failures
+            failures.append((info.name, repr(e)))
+    assert not failures, failures
