@@ -155,6 +155,37 @@ class _CdnaConv3x3Fn(torch.autograd.Function):
         return dx, dw_raw, dbias
 
 
+class _WrwConv2dFn(torch.autograd.Function):
+    """MIOpen forward + MIOpen dx, hand-written MFMA wrw (dW).
+
+    For the ResNet layer-1 family (32x32 C64->K64 s1p1, the largest-R wrw
+    shapes) MIOpen's wrw igemm cannot reuse one x tile across the 9 taps;
+    conv3x3_wrw stages the tile TRANSPOSED in LDS once (three pre-shifted
+    copies for aligned reads) and contracts all taps from it — measured
+    252 TF vs tuned MIOpen's 206 (1.23x) at batch 128
+    (profiles/kernels_summary.md)."""
+
+    @staticmethod
+    def forward(ctx, x_cl: torch.Tensor, weight: torch.Tensor):
+        w16 = weight.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        y = torch.ops.aten.convolution(x_cl, w16, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1)
+        ctx.save_for_backward(x_cl, w16)
+        ctx.w_dtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, gy: torch.Tensor):
+        x_cl, w16 = ctx.saved_tensors
+        gy_cl = gy.contiguous(memory_format=torch.channels_last)
+        dx = torch.ops.aten.convolution_backward(
+            gy_cl, x_cl, w16, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [True, False, False],
+        )[0]
+        # channels_last NCHW memory IS NHWC: both permutes are zero-copy views
+        dw = _C.conv3x3_wrw(x_cl.permute(0, 2, 3, 1), gy_cl.permute(0, 2, 3, 1))
+        return dx, dw.float() if ctx.w_dtype == torch.float32 else dw
+
+
 class CdnaConv2d(nn.Conv2d):
     """Drop-in nn.Conv2d for 3x3/s1/p1/groups=1 that runs the MFMA direct
     kernel on GPU bf16 channels-last inputs; falls back to F.conv2d off that
@@ -190,8 +221,38 @@ class CdnaConv2d(nn.Conv2d):
             input.shape[3], self.out_channels, self.in_channels
         )
 
+    def _wrw_path(self, input: torch.Tensor) -> bool:
+        """MIOpen fwd/dx + custom MFMA wrw: measured win at 32x32 C64->K64
+        with batch >= ~96 (1.23x tuned MIOpen at N=128; parity at N=64).
+        FL4_WRW=0 disables."""
+        import os
+
+        return (
+            HAS_EXT
+            and os.environ.get("FL4_WRW", "1") != "0"
+            and input.is_cuda
+            and input.dtype == torch.bfloat16
+            and self.weight.dtype == torch.bfloat16
+            and self.bias is None
+            and self.kernel_size == (3, 3)
+            and self.stride == (1, 1)
+            and self.padding == (1, 1)
+            and self.dilation == (1, 1)
+            and self.groups == 1
+            and self.padding_mode == "zeros"
+            and self.in_channels == 64
+            and self.out_channels == 64
+            and input.shape[3] == 32
+            and input.shape[2] % 4 == 0
+            and input.shape[0] >= 96
+        )
+
     def forward(self, input: torch.Tensor) -> torch.Tensor:
         if not self._fast_path(input):
+            if self._wrw_path(input):
+                return _WrwConv2dFn.apply(
+                    input.contiguous(memory_format=torch.channels_last), self.weight
+                )
             return super().forward(input)
         # channels_last NCHW memory IS NHWC: the permute+contiguous is free
         x_nhwc = input.permute(0, 2, 3, 1).contiguous()

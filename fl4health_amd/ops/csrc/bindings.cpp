@@ -35,6 +35,8 @@ void launch_conv3x3_fwd_kb32(const void*, const void*, const float*, void*, int,
 void launch_pack_kb32(const void*, void*, int, int, int, hipStream_t);
 void launch_conv3x3_fwd_kzloop(const void*, const void*, const float*, void*, int, int, int, int,
                                int, hipStream_t);
+void launch_conv3x3_wrw(const void*, const void*, void*, float*, void*, int, int, int, int, int,
+                        hipStream_t);
 void launch_coo_count(const float*, float, int64_t, int64_t, int32_t*, hipStream_t);
 void launch_coo_write(const float*, const float*, float, const int32_t*, float*, int64_t*,
                       const int64_t*, int, int64_t, int64_t, int64_t, hipStream_t);
@@ -257,6 +259,26 @@ int bn_dtype_of(const torch::Tensor& t) {
 }
 
 // x: NHWC-flattened [R, C] contiguous. Returns (y, save_mean, save_invstd).
+// wrw: dW for 3x3/s1/p1 NHWC bf16 (v1: C=K=64, W=32 - the layer-1 family).
+// Returns bf16 [K, C, 3, 3] in channels_last memory (matches what
+// aten::convolution_backward hands back for channels_last convs).
+torch::Tensor conv3x3_wrw(torch::Tensor x, torch::Tensor dy) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4, "x must be NHWC contiguous");
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 4, "dy must be NHWC contiguous");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && dy.scalar_type() == torch::kBFloat16);
+  int Nn = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2), C = (int)x.size(3);
+  int K = (int)dy.size(3);
+  TORCH_CHECK(C == 64 && K == 64 && W == 32 && H % 4 == 0, "wrw v1 supports C=K=64, W=32");
+  auto bopts = torch::TensorOptions().dtype(torch::kBFloat16).device(x.device());
+  auto partial = torch::empty({256LL * 9 * C * K}, bopts);
+  auto mid = torch::empty({16LL * 9 * C * K},
+                          torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+  auto dw = torch::empty_strided({K, C, 3, 3}, {(int64_t)9 * C, 1, (int64_t)3 * C, C}, bopts);
+  launch_conv3x3_wrw(x.data_ptr(), dy.data_ptr(), partial.data_ptr(), mid.data_ptr<float>(),
+                     dw.data_ptr(), Nn, H, W, C, K, stream());
+  return dw;
+}
+
 std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, c10::optional<torch::Tensor> gamma,
                                         c10::optional<torch::Tensor> beta,
                                         c10::optional<torch::Tensor> running_mean,
@@ -534,6 +556,7 @@ torch::Tensor mkmmd_backward(torch::Tensor d, torch::Tensor gammas, torch::Tenso
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv3x3_wrw", &conv3x3_wrw, "3x3 wrw (dW) for NHWC bf16");
   m.def("bn_fwd_train", &bn_fwd_train, "NHWC batchnorm training forward", py::arg("x"),
         py::arg("gamma"), py::arg("beta"), py::arg("running_mean"), py::arg("running_var"),
         py::arg("momentum"), py::arg("eps"), py::arg("fuse_relu"), py::arg("res"),

@@ -716,3 +716,165 @@ extern "C" void launch_conv3x3_fwd_kzloop(const void* x, const void* wimg, const
         h_groups);
   }
 }
+
+// ---------------------------------------------------------------------------
+// conv3x3 backward-WEIGHTS (wrw): dW[tap][c][k] = sum_r x[r_shift, c] dy[r, k]
+// v1 scope: NHWC bf16, stride1/pad1, C=K=64, W=32 (the ResNet layer-1 family
+// - the shapes where MIOpen's wrw igemm runs ~4x off the HBM roofline because
+// it cannot reuse one x tile across all 9 taps).
+//
+// Design: deterministic split-K over (n, h-group) tiles. Each WG stages one
+// tile TRANSPOSED in LDS - x_t[c][(BH+2) x (W+2) halo patch] and
+// dy_t[k][BH x W] - because the MFMA contraction dim here is the ROW dim:
+// both operands feed __builtin_amdgcn_mfma_f32_16x16x32_bf16 as
+// [channel][row] fragments, and one staged x patch serves all 9 taps (the
+// tap shift is just a different base offset into the patch). 8 waves = 4
+// m-groups (c) x 2 n-groups (k); acc = 2 n-tiles x 9 taps x f32x4 per wave.
+// Each WG accumulates its tile range in registers and writes ONE fp32
+// partial dW; a combine kernel reduces the fixed split count (deterministic).
+// ---------------------------------------------------------------------------
+#define WRW_BH 4
+#define WRW_SPLITS 256
+// x is staged THREE times, one copy per horizontal tap shift, so every
+// MFMA A-fragment read is a 16B-aligned ds_read_b128 (a single shifted copy
+// forces odd-u16 offsets -> scalar LDS reads, measured 0.46x MIOpen).
+// Pitches are multiples of 8 elements (alignment) whose dword stride is
+// ≡ 4 (mod 32), spreading 16 reading lanes over 8 LDS banks (2-way).
+// The transpose STORES additionally XOR-swizzle the 8-element block index
+// by (c>>3)&7: the 8 lanes of a store group share sp but differ in c8, and
+// their channel stride is ≡ 0 mod 32 banks — unswizzled they all hit one
+// bank (8-way conflict on every staging store).
+#define WRW_SPITCH 264  // per-channel pitch of one x copy (40 cols padded)
+// block-swizzled LDS offset within one x copy: channel c, element sp
+__device__ __forceinline__ size_t wrw_swz(int c, int sp) {
+  return (size_t)c * WRW_SPITCH + (((sp >> 3) ^ ((c >> 3) & 7)) << 3) + (sp & 7);
+}
+#define WRW_DPITCH 136  // per-channel pitch of dy (BH*32 = 128 cols padded)
+
+__global__ __launch_bounds__(512) void conv3x3_wrw_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ dy, bf16* __restrict__ partial,
+    int Nn, int H, int W, int C, int K, int n_tiles) {
+  __shared__ __bf16 s_xt[3 * 64 * WRW_SPITCH];
+  __shared__ __bf16 s_dyt[64 * WRW_DPITCH];
+  int tid = threadIdx.x;
+  int wave = tid >> 6, lane = tid & 63;
+  int mg = wave & 3, ng = wave >> 2;
+  int ln = lane & 15, km = lane >> 4;
+  int h_groups = H / WRW_BH;
+
+  f32x4 acc[2][9];
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) acc[t][tap] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int tile = blockIdx.x; tile < n_tiles; tile += WRW_SPLITS) {
+    int n0 = tile / h_groups;
+    int h0 = (tile % h_groups) * WRW_BH;
+    // stage x: lines h0-1 .. h0+BH, w -1..W; copy d holds x[w = col + d - 1]
+    // at col = w + 1 - d (cols 0..39 inside a 40-col line)
+    const int xtotal = (WRW_BH + 2) * 34 * (64 / 8);
+    for (int idx = tid; idx < xtotal; idx += 512) {
+      int c8 = idx & 7;
+      int sp = idx >> 3;
+      int line = sp / 34, wx = sp % 34;
+      int hh = h0 - 1 + line, ww = wx - 1;
+      bf16x8 v = bf16x8{};
+      if (hh >= 0 && hh < H && ww >= 0 && ww < W)
+        v = *reinterpret_cast<const bf16x8*>(
+            &x[(((int64_t)n0 * H + hh) * W + ww) * C + c8 * 8]);
+#pragma unroll
+      for (int d = 0; d < 3; ++d) {
+        int col = ww + 1 - d;  // copy d: col w holds x[w + d - 1]
+        if (col < 0 || col >= 40) continue;
+        __bf16* dst = &s_xt[((size_t)d * 64) * WRW_SPITCH];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dst[wrw_swz(c8 * 8 + j, line * 40 + col)] = v[j];
+      }
+    }
+    const int dtotal = WRW_BH * 32 * (64 / 8);
+    for (int idx = tid; idx < dtotal; idx += 512) {
+      int k8 = idx & 7;
+      int sp = idx >> 3;
+      int line = sp >> 5, ww = sp & 31;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          &dy[(((int64_t)n0 * H + h0 + line) * W + ww) * K + k8 * 8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s_dyt[(size_t)(k8 * 8 + j) * WRW_DPITCH + sp] = v[j];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < WRW_BH; ++j) {
+      bf16x8 bfrag[2];
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+        bfrag[t] = *reinterpret_cast<const bf16x8*>(
+            &s_dyt[(size_t)((ng * 2 + t) * 16 + ln) * WRW_DPITCH + j * 32 + km * 8]);
+#pragma unroll
+      for (int tap = 0; tap < 9; ++tap) {
+        int dyy = tap / 3, dxx = tap % 3;
+        bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+            &s_xt[(size_t)dxx * 64 * WRW_SPITCH + wrw_swz(mg * 16 + ln, (j + dyy) * 40 + km * 8)]);
+#pragma unroll
+        for (int t = 0; t < 2; ++t)
+          acc[t][tap] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag[t], acc[t][tap], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  // partial[split][tap][c][k] bf16 (fp32 accumulated in regs, rounded once)
+  bf16* base = partial + (int64_t)blockIdx.x * 9 * C * K;
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int c = mg * 16 + km * 4 + r;
+        int k = (ng * 2 + t) * 16 + ln;
+        base[((int64_t)tap * C + c) * K + k] = (bf16)acc[t][tap][r];
+      }
+}
+
+// Two-stage split combine. A single 144-block pass serially reading all 256
+// strided splits left most of the chip idle (~25 us — dominated the call);
+// stage 1 fans the split dim across gridDim.y, stage 2 folds the remainder.
+#define WRW_SG 16  // split groups in stage 1
+__global__ __launch_bounds__(256) void conv3x3_wrw_combine1_kernel(
+    const bf16* __restrict__ partial, float* __restrict__ mid, int total, int splits) {
+  int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= total) return;
+  int per = splits / WRW_SG;
+  float s = 0.f;
+  for (int sp = blockIdx.y * per; sp < (blockIdx.y + 1) * per; ++sp)
+    s += (float)partial[(int64_t)sp * total + idx];
+  mid[(int64_t)blockIdx.y * total + idx] = s;
+}
+
+__global__ __launch_bounds__(256) void conv3x3_wrw_combine2_kernel(
+    const float* __restrict__ mid, bf16* __restrict__ dw_cl, int C, int K) {
+  int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  int total = 9 * C * K;
+  if (idx >= total) return;
+  float s = 0.f;
+#pragma unroll
+  for (int g = 0; g < WRW_SG; ++g) s += mid[(int64_t)g * total + idx];
+  int tap = idx / (C * K);
+  int c = (idx / K) % C;
+  int k = idx % K;
+  dw_cl[((int64_t)k * 9 + tap) * C + c] = (bf16)s;
+}
+
+extern "C" void launch_conv3x3_wrw(const void* x, const void* dy, void* partial, float* mid,
+                                   void* dw_cl, int Nn, int H, int W, int C, int K,
+                                   hipStream_t s) {
+  int n_tiles = Nn * (H / WRW_BH);
+  conv3x3_wrw_kernel<<<dim3(WRW_SPLITS, 1, 1), 512, 0, s>>>(
+      (const bf16*)x, (const bf16*)dy, (bf16*)partial, Nn, H, W, C, K, n_tiles);
+  int total = 9 * C * K;
+  dim3 g1((total + 255) / 256, WRW_SG, 1);
+  conv3x3_wrw_combine1_kernel<<<g1, 256, 0, s>>>((const bf16*)partial, mid, total, WRW_SPLITS);
+  conv3x3_wrw_combine2_kernel<<<dim3((total + 255) / 256, 1, 1), 256, 0, s>>>(
+      mid, (bf16*)dw_cl, C, K);
+}
