@@ -217,3 +217,33 @@ def test_masked_normalization_layers_roundtrip():
     assert all("score" in n for n in trainable)
     out.sum().backward()
     assert all(p.grad is not None for n, p in masked.named_parameters() if p.requires_grad)
+
+
+def test_compute_dice_on_count_tensors_drop_and_replace():
+    from fl4health_amd.metrics.metrics_utils import compute_dice_on_count_tensors
+
+    tp = torch.tensor([2.0, 0.0, 0.0])
+    fp = torch.tensor([1.0, 0.0, 2.0])
+    fn = torch.tensor([1.0, 0.0, 0.0])
+    # entry 1 is all-true-negative (undefined)
+    dropped = compute_dice_on_count_tensors(tp, fp, fn, zero_division=None)
+    assert dropped.shape == (2,)
+    assert dropped[0] == pytest.approx(4.0 / 6.0)
+    assert dropped[1] == pytest.approx(0.0)
+    replaced = compute_dice_on_count_tensors(tp, fp, fn, zero_division=1.0)
+    assert replaced.shape == (3,)
+    assert replaced[1] == pytest.approx(1.0)
+
+
+def test_threshold_tensor_float_and_label_dim():
+    from fl4health_amd.metrics.metrics_utils import threshold_tensor
+
+    x = torch.tensor([[0.2, 0.8], [0.9, 0.1]])
+    hard = threshold_tensor(x, 0.5)
+    assert torch.equal(hard, torch.tensor([[0.0, 1.0], [1.0, 0.0]]))
+    onehot = threshold_tensor(x, 1)  # argmax along dim 1
+    assert torch.equal(onehot, torch.tensor([[0.0, 1.0], [1.0, 0.0]]))
+    with pytest.raises(ValueError):
+        threshold_tensor(x, 5)
+    with pytest.raises(ValueError):
+        threshold_tensor(x, "bad")  # type: ignore[arg-type]
