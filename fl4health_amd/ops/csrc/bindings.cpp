@@ -259,7 +259,8 @@ int bn_dtype_of(const torch::Tensor& t) {
 }
 
 // x: NHWC-flattened [R, C] contiguous. Returns (y, save_mean, save_invstd).
-// wrw: dW for 3x3/s1/p1 NHWC bf16 (v1: C=K=64, W=32 - the layer-1 family).
+// wrw: dW for 3x3/s1/p1 NHWC bf16 (C,K in {64,128}; W=32 or 16 - the ResNet
+// layer-1/layer-2 families; >64-channel cases run as 64x64 sub-slices).
 // Returns bf16 [K, C, 3, 3] in channels_last memory (matches what
 // aten::convolution_backward hands back for channels_last convs).
 torch::Tensor conv3x3_wrw(torch::Tensor x, torch::Tensor dy) {
@@ -268,10 +269,12 @@ torch::Tensor conv3x3_wrw(torch::Tensor x, torch::Tensor dy) {
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && dy.scalar_type() == torch::kBFloat16);
   int Nn = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2), C = (int)x.size(3);
   int K = (int)dy.size(3);
-  TORCH_CHECK(C == 64 && K == 64 && W == 32 && H % 4 == 0, "wrw v1 supports C=K=64, W=32");
+  TORCH_CHECK((C == 64 || C == 128) && (K == 64 || K == 128), "wrw supports C,K in {64,128}");
+  TORCH_CHECK((W == 32 && H % 4 == 0) || (W == 16 && H % 8 == 0), "wrw supports W=32 (BH 4) or W=16 (BH 8)");
   auto bopts = torch::TensorOptions().dtype(torch::kBFloat16).device(x.device());
-  auto partial = torch::empty({256LL * 9 * C * K}, bopts);
-  auto mid = torch::empty({16LL * 9 * C * K},
+  // one 64x64 sub-slice at a time: partial/mid are reused across sub-launches
+  auto partial = torch::empty({256LL * 9 * 64 * 64}, bopts);
+  auto mid = torch::empty({16LL * 9 * 64 * 64},
                           torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
   auto dw = torch::empty_strided({K, C, 3, 3}, {(int64_t)9 * C, 1, (int64_t)3 * C, C}, bopts);
   launch_conv3x3_wrw(x.data_ptr(), dy.data_ptr(), partial.data_ptr(), mid.data_ptr<float>(),

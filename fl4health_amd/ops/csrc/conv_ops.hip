@@ -733,7 +733,6 @@ extern "C" void launch_conv3x3_fwd_kzloop(const void* x, const void* wimg, const
 // Each WG accumulates its tile range in registers and writes ONE fp32
 // partial dW; a combine kernel reduces the fixed split count (deterministic).
 // ---------------------------------------------------------------------------
-#define WRW_BH 4
 #define WRW_SPLITS 256
 // x is staged THREE times, one copy per horizontal tap shift, so every
 // MFMA A-fragment read is a 16B-aligned ds_read_b128 (a single shifted copy
@@ -744,23 +743,32 @@ extern "C" void launch_conv3x3_fwd_kzloop(const void* x, const void* wimg, const
 // by (c>>3)&7: the 8 lanes of a store group share sp but differ in c8, and
 // their channel stride is ≡ 0 mod 32 banks — unswizzled they all hit one
 // bank (8-way conflict on every staging store).
-#define WRW_SPITCH 264  // per-channel pitch of one x copy (40 cols padded)
-// block-swizzled LDS offset within one x copy: channel c, element sp
+// Index math for both geometries is bit-verified by the host simulator that
+// produced this layout (see commit history / tools/wrw_micro.py).
+#define WRW_SPITCH 264
+#define WRW_DPITCH 136
 __device__ __forceinline__ size_t wrw_swz(int c, int sp) {
   return (size_t)c * WRW_SPITCH + (((sp >> 3) ^ ((c >> 3) & 7)) << 3) + (sp & 7);
 }
-#define WRW_DPITCH 136  // per-channel pitch of dy (BH*32 = 128 cols padded)
 
+// One 64-channel x 64-filter wrw slice: x[...,c0:c0+64] against
+// dy[...,k0:k0+64] with physical strides XC/KC (C/K up to 128 run as 2x2
+// sub-slices from the launcher — LDS cannot hold three shifted 128-channel
+// copies). W_=32/BH_=4 covers ResNet layer 1, W_=16/BH_=8 layer 2; a
+// 32-row MFMA k-chunk spans 32/W_ image lines and each lane's 8 contiguous
+// rows stay within one line for W_ >= 8.
+template <int W_, int BH_>
 __global__ __launch_bounds__(512) void conv3x3_wrw_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ dy, bf16* __restrict__ partial,
-    int Nn, int H, int W, int C, int K, int n_tiles) {
+    int Nn, int H, int XC, int c0, int KC, int k0, int n_tiles) {
+  constexpr int LP = W_ + 8;  // staged line pitch (W_+2 halo cols, padded)
   __shared__ __bf16 s_xt[3 * 64 * WRW_SPITCH];
   __shared__ __bf16 s_dyt[64 * WRW_DPITCH];
   int tid = threadIdx.x;
   int wave = tid >> 6, lane = tid & 63;
   int mg = wave & 3, ng = wave >> 2;
   int ln = lane & 15, km = lane >> 4;
-  int h_groups = H / WRW_BH;
+  int h_groups = H / BH_;
 
   f32x4 acc[2][9];
 #pragma unroll
@@ -770,52 +778,53 @@ __global__ __launch_bounds__(512) void conv3x3_wrw_kernel(
 
   for (int tile = blockIdx.x; tile < n_tiles; tile += WRW_SPLITS) {
     int n0 = tile / h_groups;
-    int h0 = (tile % h_groups) * WRW_BH;
-    // stage x: lines h0-1 .. h0+BH, w -1..W; copy d holds x[w = col + d - 1]
-    // at col = w + 1 - d (cols 0..39 inside a 40-col line)
-    const int xtotal = (WRW_BH + 2) * 34 * (64 / 8);
+    int h0 = (tile % h_groups) * BH_;
+    // stage x: lines h0-1 .. h0+BH_, w -1..W_; copy d holds x[w = col + d - 1]
+    const int xtotal = (BH_ + 2) * (W_ + 2) * (64 / 8);
     for (int idx = tid; idx < xtotal; idx += 512) {
       int c8 = idx & 7;
       int sp = idx >> 3;
-      int line = sp / 34, wx = sp % 34;
+      int line = sp / (W_ + 2), wx = sp % (W_ + 2);
       int hh = h0 - 1 + line, ww = wx - 1;
       bf16x8 v = bf16x8{};
-      if (hh >= 0 && hh < H && ww >= 0 && ww < W)
+      if (hh >= 0 && hh < H && ww >= 0 && ww < W_)
         v = *reinterpret_cast<const bf16x8*>(
-            &x[(((int64_t)n0 * H + hh) * W + ww) * C + c8 * 8]);
+            &x[(((int64_t)n0 * H + hh) * W_ + ww) * XC + c0 + c8 * 8]);
 #pragma unroll
       for (int d = 0; d < 3; ++d) {
         int col = ww + 1 - d;  // copy d: col w holds x[w + d - 1]
-        if (col < 0 || col >= 40) continue;
+        if (col < 0 || col >= LP) continue;
         __bf16* dst = &s_xt[((size_t)d * 64) * WRW_SPITCH];
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          dst[wrw_swz(c8 * 8 + j, line * 40 + col)] = v[j];
+          dst[wrw_swz(c8 * 8 + j, line * LP + col)] = v[j];
       }
     }
-    const int dtotal = WRW_BH * 32 * (64 / 8);
+    const int dtotal = BH_ * W_ * (64 / 8);
     for (int idx = tid; idx < dtotal; idx += 512) {
       int k8 = idx & 7;
       int sp = idx >> 3;
-      int line = sp >> 5, ww = sp & 31;
+      int line = sp / W_, ww = sp % W_;
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
-          &dy[(((int64_t)n0 * H + h0 + line) * W + ww) * K + k8 * 8]);
+          &dy[(((int64_t)n0 * H + h0 + line) * W_ + ww) * KC + k0 + k8 * 8]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) s_dyt[(size_t)(k8 * 8 + j) * WRW_DPITCH + sp] = v[j];
     }
     __syncthreads();
 #pragma unroll
-    for (int j = 0; j < WRW_BH; ++j) {
+    for (int kc = 0; kc < BH_ * W_ / 32; ++kc) {
+      int r0 = kc * 32 + km * 8;
+      int line = r0 / W_, col = r0 % W_;
       bf16x8 bfrag[2];
 #pragma unroll
       for (int t = 0; t < 2; ++t)
         bfrag[t] = *reinterpret_cast<const bf16x8*>(
-            &s_dyt[(size_t)((ng * 2 + t) * 16 + ln) * WRW_DPITCH + j * 32 + km * 8]);
+            &s_dyt[(size_t)((ng * 2 + t) * 16 + ln) * WRW_DPITCH + r0]);
 #pragma unroll
       for (int tap = 0; tap < 9; ++tap) {
         int dyy = tap / 3, dxx = tap % 3;
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-            &s_xt[(size_t)dxx * 64 * WRW_SPITCH + wrw_swz(mg * 16 + ln, (j + dyy) * 40 + km * 8)]);
+            &s_xt[(size_t)dxx * 64 * WRW_SPITCH + wrw_swz(mg * 16 + ln, (line + dyy) * LP + col)]);
 #pragma unroll
         for (int t = 0; t < 2; ++t)
           acc[t][tap] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag[t], acc[t][tap], 0, 0, 0);
@@ -824,7 +833,7 @@ __global__ __launch_bounds__(512) void conv3x3_wrw_kernel(
     __syncthreads();
   }
   // partial[split][tap][c][k] bf16 (fp32 accumulated in regs, rounded once)
-  bf16* base = partial + (int64_t)blockIdx.x * 9 * C * K;
+  bf16* base = partial + (int64_t)blockIdx.x * 9 * 64 * 64;
 #pragma unroll
   for (int t = 0; t < 2; ++t)
 #pragma unroll
@@ -833,13 +842,14 @@ __global__ __launch_bounds__(512) void conv3x3_wrw_kernel(
       for (int r = 0; r < 4; ++r) {
         int c = mg * 16 + km * 4 + r;
         int k = (ng * 2 + t) * 16 + ln;
-        base[((int64_t)tap * C + c) * K + k] = (bf16)acc[t][tap][r];
+        base[((int64_t)tap * 64 + c) * 64 + k] = (bf16)acc[t][tap][r];
       }
 }
 
 // Two-stage split combine. A single 144-block pass serially reading all 256
 // strided splits left most of the chip idle (~25 us — dominated the call);
-// stage 1 fans the split dim across gridDim.y, stage 2 folds the remainder.
+// stage 1 fans the split dim across gridDim.y, stage 2 folds the remainder
+// and scatters the 64x64 slice into the full [K,C,3,3] channels_last dW.
 #define WRW_SG 16  // split groups in stage 1
 __global__ __launch_bounds__(256) void conv3x3_wrw_combine1_kernel(
     const bf16* __restrict__ partial, float* __restrict__ mid, int total, int splits) {
@@ -853,28 +863,38 @@ __global__ __launch_bounds__(256) void conv3x3_wrw_combine1_kernel(
 }
 
 __global__ __launch_bounds__(256) void conv3x3_wrw_combine2_kernel(
-    const float* __restrict__ mid, bf16* __restrict__ dw_cl, int C, int K) {
+    const float* __restrict__ mid, bf16* __restrict__ dw_cl, int CT, int KT, int c0, int k0) {
   int idx = blockIdx.x * blockDim.x + threadIdx.x;
-  int total = 9 * C * K;
+  const int total = 9 * 64 * 64;
   if (idx >= total) return;
   float s = 0.f;
 #pragma unroll
   for (int g = 0; g < WRW_SG; ++g) s += mid[(int64_t)g * total + idx];
-  int tap = idx / (C * K);
-  int c = (idx / K) % C;
-  int k = idx % K;
-  dw_cl[((int64_t)k * 9 + tap) * C + c] = (bf16)s;
+  int tap = idx / (64 * 64);
+  int c = (idx / 64) % 64;
+  int k = idx % 64;
+  dw_cl[((int64_t)(k0 + k) * 9 + tap) * CT + c0 + c] = (bf16)s;
 }
 
 extern "C" void launch_conv3x3_wrw(const void* x, const void* dy, void* partial, float* mid,
                                    void* dw_cl, int Nn, int H, int W, int C, int K,
                                    hipStream_t s) {
-  int n_tiles = Nn * (H / WRW_BH);
-  conv3x3_wrw_kernel<<<dim3(WRW_SPLITS, 1, 1), 512, 0, s>>>(
-      (const bf16*)x, (const bf16*)dy, (bf16*)partial, Nn, H, W, C, K, n_tiles);
-  int total = 9 * C * K;
+  const int total = 9 * 64 * 64;
   dim3 g1((total + 255) / 256, WRW_SG, 1);
-  conv3x3_wrw_combine1_kernel<<<g1, 256, 0, s>>>((const bf16*)partial, mid, total, WRW_SPLITS);
-  conv3x3_wrw_combine2_kernel<<<dim3((total + 255) / 256, 1, 1), 256, 0, s>>>(
-      mid, (bf16*)dw_cl, C, K);
+  dim3 g2((total + 255) / 256, 1, 1);
+  for (int c0 = 0; c0 < C; c0 += 64)
+    for (int k0 = 0; k0 < K; k0 += 64) {
+      if (W == 32) {
+        int n_tiles = Nn * (H / 4);
+        conv3x3_wrw_kernel<32, 4><<<dim3(WRW_SPLITS, 1, 1), 512, 0, s>>>(
+            (const bf16*)x, (const bf16*)dy, (bf16*)partial, Nn, H, C, c0, K, k0, n_tiles);
+      } else {
+        int n_tiles = Nn * (H / 8);
+        conv3x3_wrw_kernel<16, 8><<<dim3(WRW_SPLITS, 1, 1), 512, 0, s>>>(
+            (const bf16*)x, (const bf16*)dy, (bf16*)partial, Nn, H, C, c0, K, k0, n_tiles);
+      }
+      conv3x3_wrw_combine1_kernel<<<g1, 256, 0, s>>>((const bf16*)partial, mid, total,
+                                                     WRW_SPLITS);
+      conv3x3_wrw_combine2_kernel<<<g2, 256, 0, s>>>(mid, (bf16*)dw_cl, C, K, c0, k0);
+    }
 }

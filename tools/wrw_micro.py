@@ -49,3 +49,6 @@ def run(n, h, w, c, k):
 if __name__ == "__main__":
     run(128, 32, 32, 64, 64)
     run(64, 32, 32, 64, 64)
+    run(128, 16, 16, 128, 128)
+    run(128, 16, 16, 64, 128)
+    run(128, 16, 16, 128, 64)
