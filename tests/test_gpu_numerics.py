@@ -675,9 +675,30 @@ def test_conv3x3_kzloop_matches_reference():
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("h,w,c,k", [(32, 32, 64, 64), (16, 16, 128, 128), (16, 16, 64, 128)])
+def test_conv3x3_wrw_kernel_numerics(h, w, c, k):
+    """Custom MFMA wrw (dW) kernel vs fp32 autograd reference across the
+    templated geometries (W32/BH4 adopted; W16/BH8 correct but unadopted)."""
+    import torch.nn.functional as Fn
+
+    from fl4health_amd import _C
+
+    torch.manual_seed(0)
+    n = 32
+    x = (torch.randn(n, h, w, c, device="cuda") * 0.5).to(torch.bfloat16)
+    dy = (torch.randn(n, h, w, k, device="cuda") * 0.5).to(torch.bfloat16)
+    dw = _C.conv3x3_wrw(x, dy)
+    xf = x.permute(0, 3, 1, 2).float().requires_grad_(True)
+    wf = torch.zeros(k, c, 3, 3, device="cuda", requires_grad=True)
+    Fn.conv2d(xf, wf, None, 1, 1).backward(dy.permute(0, 3, 1, 2).float())
+    rel = (dw.float() - wf.grad).norm() / wf.grad.norm()
+    assert rel < 2e-2, f"wrw numerics off ({h}x{w} C{c}K{k}): rel={rel}"
+    assert torch.equal(dw, _C.conv3x3_wrw(x, dy))  # deterministic split-K
+
+
+@pytest.mark.gpu
 def test_conv3x3_wrw_matches_reference():
-    """Custom MFMA wrw (dW) kernel vs fp32 autograd reference, plus the
-    autograd adoption path (_WrwConv2dFn / CdnaConv2d._wrw_path)."""
+    """The autograd adoption path (_WrwConv2dFn / CdnaConv2d._wrw_path)."""
     import torch.nn.functional as Fn
 
     from fl4health_amd import _C
