@@ -10,7 +10,10 @@ from __future__ import annotations
 
 from typing import Any
 
+import logging
+
 from fl4health_amd.client_managers.base import ClientProxy
+log = logging.getLogger(__name__)
 from fl4health_amd.common import (
     EvaluateIns,
     EvaluateRes,
@@ -62,6 +65,7 @@ class InProcessTransport:
             try:
                 results.append((proxy, proxy.fit(ins, timeout)))
             except Exception as e:  # noqa: BLE001 - client failure policy
+                log.exception("client %s fit failed (accept_failures=%s)", proxy.cid, self.accept_failures)
                 if not self.accept_failures:
                     raise
                 failures.append(e)
@@ -74,6 +78,7 @@ class InProcessTransport:
             try:
                 results.append((proxy, proxy.evaluate(ins, timeout)))
             except Exception as e:  # noqa: BLE001
+                log.exception("client %s evaluate failed (accept_failures=%s)", proxy.cid, self.accept_failures)
                 if not self.accept_failures:
                     raise
                 failures.append(e)
