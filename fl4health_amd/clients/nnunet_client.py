@@ -30,6 +30,7 @@ import gc
 import importlib.util
 import logging
 import pickle
+import zlib
 
 import torch
 from torch.utils.data import DataLoader
@@ -97,7 +98,7 @@ class NnunetClient(BasicClient):
         ncls = int(config.get("num_classes", 3))
         lo = int(config.get("min_volume_size", 28))
         hi = int(config.get("max_volume_size", 44))
-        gen = torch.Generator().manual_seed(abs(hash(self.client_name)) % (2**31))
+        gen = torch.Generator().manual_seed(zlib.crc32(self.client_name.encode()) % (2**31))
         vols, segs, spacings = [], [], []
         for _ in range(n):
             shape = [int(torch.randint(lo, hi, (), generator=gen)) for _ in range(3)]
@@ -257,13 +258,13 @@ class NnunetClient(BasicClient):
             pv[:n_train], ps[:n_train], cfg["patch_size"], int(cfg["batch_size"]),
             n_batches_per_epoch=int(config.get("n_batches_per_epoch", 10)),
             deep_supervision_levels=net["num_levels"],
-            seed=abs(hash(self.client_name)) % (2**31),
+            seed=zlib.crc32(self.client_name.encode()) % (2**31),
         )
         val_aug = AsyncPatchLoader(
             pv[n_train:] or pv[:1], ps[n_train:] or ps[:1], cfg["patch_size"], int(cfg["batch_size"]),
             n_batches_per_epoch=int(config.get("n_val_batches", 2)),
             deep_supervision_levels=1,
-            seed=1 + abs(hash(self.client_name)) % (2**31),
+            seed=1 + zlib.crc32(self.client_name.encode()) % (2**31),
         )
         train = NnUNetDataLoaderWrapper(train_aug, self.nnunet_config)
         val = NnUNetDataLoaderWrapper(val_aug, self.nnunet_config)

@@ -128,9 +128,14 @@ def plan_experiment(
     # clip the patch to the voxel budget, shrinking the largest axis first
     while np.prod(patch) > max_patch_voxels:
         patch[int(np.argmax(patch))] = math.ceil(patch[int(np.argmax(patch))] * 0.9)
-    # pooling depth per axis + divisor rounding
-    n_levels = int(min(max_levels, max(2, math.floor(math.log2(max(np.min(patch), 4))))))
-    div = 2 ** (n_levels - 1)
+    # pooling depth + divisor rounding. UNet3D pools after EVERY encoder
+    # level (num_levels poolings before the bottleneck), so each patch dim
+    # must divide by 2**n_levels — 2**(n_levels-1) left dims like 14 at two
+    # levels, and the decoder cat then failed on the 7-voxel skip (a
+    # PYTHONHASHSEED-dependent flake: synthetic volume sizes ride
+    # hash(client_name) seeds).
+    n_levels = int(min(max_levels, max(2, math.floor(math.log2(max(np.min(patch), 4))) - 1)))
+    div = 2 ** n_levels
     patch_size = [max(div, int(round(p / div) * div)) for p in patch]
     n_train = int(dataset_json.get("numTraining", len(resampled)))
     max_voxels = float(np.prod(median_shape)) * n_train * 0.05
