@@ -169,3 +169,24 @@ def test_poly_lr_wrapper_decay():
         opt.step()
         sched.step()
     assert lrs[0] == 1.0 and all(a > b for a, b in zip(lrs, lrs[1:]))
+
+
+def test_plan_patch_divisible_by_pooling_factor():
+    """UNet3D pools after EVERY encoder level: every patch dim the planner
+    emits must divide by 2**n_stages (regression for the 14-voxel decoder-cat
+    crash; awkward odd/even-but-not-4 shapes all round cleanly)."""
+    from fl4health_amd.preprocessing.nnunet import plan_experiment
+
+    for sizes in [[14, 17, 19], [9, 9, 9], [15, 20, 14], [33, 18, 10]]:
+        fp = {
+            "shapes_after_crop": [sizes, [s + 1 for s in sizes]],
+            "spacings": [[1.0, 1.0, 1.0]] * 2,
+            "foreground_intensity_properties_per_channel": {"0": {"mean": 0.0, "std": 1.0}},
+        }
+        dj = {"numTraining": 2, "channel_names": {"0": "ch"}, "labels": {"background": 0, "fg": 1}}
+        for max_levels in (2, 3, 5):
+            plans = plan_experiment(fp, dj, max_patch_voxels=16**3, max_levels=max_levels)
+            cfg = plans["configurations"]["3d_fullres"]
+            div = 2 ** cfg["n_stages"]
+            assert all(p % div == 0 for p in cfg["patch_size"]), (sizes, max_levels, cfg)
+            assert all(p >= div for p in cfg["patch_size"])
