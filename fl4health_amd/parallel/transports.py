@@ -8,12 +8,10 @@ order, so seeded runs are bit-reproducible.
 """
 from __future__ import annotations
 
+import logging
 from typing import Any
 
-import logging
-
 from fl4health_amd.client_managers.base import ClientProxy
-log = logging.getLogger(__name__)
 from fl4health_amd.common import (
     EvaluateIns,
     EvaluateRes,
@@ -24,6 +22,8 @@ from fl4health_amd.common import (
     GetPropertiesIns,
     GetPropertiesRes,
 )
+
+log = logging.getLogger(__name__)
 
 
 class InProcessClientProxy(ClientProxy):
