@@ -175,3 +175,21 @@ def test_grad_as_1d_aliases_memory_not_copies():
     # plain contiguous
     p = torch.randn(7)
     assert BasicClient._grad_as_1d(p).data_ptr() == p.data_ptr()
+
+
+def test_graph_capture_gates_custom_train_steps():
+    """Only the stock BasicClient.train_step opts into the steal-then-pack
+    grad flow under capture; subclasses with their own train_step (APFL,
+    Ditto, ensemble, flexible...) must keep their logic."""
+    from fl4health_amd.clients.apfl_client import ApflClient
+    from fl4health_amd.clients.basic_client import BasicClient
+    from fl4health_amd.clients.ditto_client import DittoClient
+    from fl4health_amd.clients.ensemble_client import EnsembleClient
+    from fl4health_amd.clients.adaptive_drift_constraint_client import FedProxClient
+    from fl4health_amd.clients.scaffold_client import ScaffoldClient
+
+    stock = lambda cls: cls.train_step is BasicClient.train_step  # noqa: E731
+    # FedProx/SCAFFOLD inherit the stock step (bench + scaffold use the fast path)
+    assert stock(FedProxClient) and stock(ScaffoldClient)
+    # personalized/ensemble clients define their own and must be gated out
+    assert not stock(ApflClient) and not stock(DittoClient) and not stock(EnsembleClient)
