@@ -1,6 +1,7 @@
 """GEMINI experiment harness (capability of reference research/gemini/:
 hospital EHR mortality/delirium prediction across 7 hospital sites with
-central / fedavg / apfl / fenda arms; the GEMINI dataset is private, so the
+central / local / fedavg / fedprox / scaffold / fedopt / fedper / ditto /
+apfl / fenda / moon / perfcl arms; the GEMINI dataset is private, so the
 harness synthesizes EHR-shaped tabular data with per-hospital covariate
 shift and picks up real site tensors from --data_dir when present).
 
@@ -19,7 +20,10 @@ from fl4health_amd.clients.adaptive_drift_constraint_client import FedProxClient
 from fl4health_amd.clients.apfl_client import ApflClient
 from fl4health_amd.clients.basic_client import BasicClient
 from fl4health_amd.clients.ditto_client import DittoClient
+from fl4health_amd.clients.fedper_client import FedPerClient
 from fl4health_amd.clients.fenda_client import FendaClient
+from fl4health_amd.clients.moon_client import MoonClient
+from fl4health_amd.clients.perfcl_client import PerFclClient
 from fl4health_amd.clients.scaffold_client import ScaffoldClient
 from fl4health_amd.common import Parameters
 from fl4health_amd.optimizers import FlatProxSGD, FlatScaffoldSGD
@@ -29,14 +33,21 @@ from fl4health_amd.strategies.scaffold import Scaffold
 from fl4health_amd.metrics.metrics import Accuracy, RocAuc
 from fl4health_amd.model_bases.apfl_base import ApflModule
 from fl4health_amd.model_bases.fenda_base import FendaModel
+from fl4health_amd.model_bases.moon_base import MoonModel
+from fl4health_amd.model_bases.perfcl_base import PerFclModel
+from fl4health_amd.model_bases.sequential_split_models import SequentiallySplitExchangeBaseModel
 from fl4health_amd.model_bases.parallel_split_models import ParallelFeatureJoinMode, ParallelSplitHeadModule
 from fl4health_amd.servers.base_server import FlServer
 from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
 from fl4health_amd.strategies.fedavg_dynamic_layer import FedAvgDynamicLayer
+from fl4health_amd.strategies.fedopt import FedAdam
 from research.common import research_argparser, run_and_record
 
 N_FEATURES = 35  # GEMINI-shaped lab/vitals feature count
-ALGORITHMS = ("fedavg", "fedprox", "scaffold", "ditto", "apfl", "fenda", "local")
+ALGORITHMS = (
+    "fedavg", "fedprox", "scaffold", "ditto", "apfl", "fenda", "local",
+    "central", "fedopt", "fedper", "moon", "perfcl",
+)
 
 
 def make_trunk() -> nn.Module:
@@ -81,9 +92,11 @@ def load_site(data_dir: str, site: int, batch_size: int, n_train: int, seed: int
 
 def build(args, device: str):
     base_cls = {
-        "fedavg": BasicClient, "local": BasicClient, "fedprox": FedProxClient,
+        "fedavg": BasicClient, "local": BasicClient, "central": BasicClient,
+        "fedopt": BasicClient, "fedprox": FedProxClient,
         "scaffold": ScaffoldClient, "ditto": DittoClient, "apfl": ApflClient,
-        "fenda": FendaClient,
+        "fenda": FendaClient, "fedper": FedPerClient, "moon": MoonClient,
+        "perfcl": PerFclClient,
     }[args.algorithm]
 
     class Client(base_cls):
@@ -96,9 +109,28 @@ def build(args, device: str):
                 return ApflModule(make_model(), adaptive_alpha=True)
             if args.algorithm == "fenda":
                 return FendaModel(make_trunk(), make_trunk(), _Head())
+            if args.algorithm == "perfcl":
+                return PerFclModel(make_trunk(), make_trunk(), _Head())
+            if args.algorithm == "fedper":
+                # trunk federated, classification head stays local
+                return SequentiallySplitExchangeBaseModel(make_trunk(), nn.Linear(32, 2))
+            if args.algorithm == "moon":
+                return MoonModel(make_trunk(), nn.Linear(32, 2))
             return make_model()
 
         def get_data_loaders(self, config):
+            if args.algorithm == "central":
+                # pooled-data baseline: one client trains on every site
+                loaders = [
+                    load_site(args.data_dir, i, args.batch_size, args.n_train, args.seed)
+                    for i in range(args.n_clients)
+                ]
+                train = torch.utils.data.ConcatDataset([dl[0].dataset for dl in loaders])
+                val = torch.utils.data.ConcatDataset([dl[1].dataset for dl in loaders])
+                return (
+                    DataLoader(train, batch_size=args.batch_size, shuffle=True),
+                    DataLoader(val, batch_size=args.batch_size),
+                )
             return load_site(args.data_dir, self.site, args.batch_size, args.n_train, args.seed)
 
         def get_criterion(self, config):
@@ -123,12 +155,16 @@ def build(args, device: str):
             if args.algorithm == "ditto" and self.optimizers.get("global") is None:
                 self.optimizers["global"] = FlatProxSGD(self.global_flat_view, lr=args.lr)
 
-    n = 1 if args.algorithm == "local" else args.n_clients  # "local": single-site baseline
+    # "local": single-site baseline; "central": one client on pooled data
+    n = 1 if args.algorithm in ("local", "central") else args.n_clients
     clients = [Client(i, metrics=[Accuracy(), RocAuc()], device=device) for i in range(n)]
     fit_cfg = lambda r: {"current_server_round": r, "local_steps": args.local_steps}  # noqa: E731
     init = Parameters([FlatParameterView(make_model()).flat.clone()])
-    if args.algorithm == "fenda":
+    if args.algorithm in ("fenda", "perfcl", "fedper"):
+        # partial-layer exchange arms aggregate per layer name
         strategy = FedAvgDynamicLayer(on_fit_config_fn=fit_cfg)
+    elif args.algorithm == "fedopt":
+        strategy = FedAdam(initial_parameters=init, on_fit_config_fn=fit_cfg)
     elif args.algorithm in ("fedprox", "ditto"):
         strategy = FedAvgWithAdaptiveConstraint(
             initial_parameters=init, initial_loss_weight=args.mu, on_fit_config_fn=fit_cfg

@@ -86,7 +86,7 @@ def test_new_research_harnesses_run(module, extra):
     assert '"algorithm"' in out.stdout
 
 
-@pytest.mark.parametrize("alg", ["apfl", "scaffold", "local"])
+@pytest.mark.parametrize("alg", ["apfl", "scaffold", "local", "central", "fedopt", "fedper", "moon", "perfcl"])
 def test_gemini_harness_runs(alg):
     env = dict(os.environ, PYTHONPATH=str(ROOT))
     out = subprocess.run(
