@@ -20,7 +20,10 @@ from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
 from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
 from research.common import research_argparser, run_and_record
 
-ALGORITHMS = ("fedavg", "ditto", "mr_mtl")
+# "fl_nnunet" is the reference's name for the federated nnU-Net arm (alias
+# of fedavg here — this harness is nnU-Net end to end); "central" trains one
+# client locally (single-node baseline).
+ALGORITHMS = ("fedavg", "fl_nnunet", "central", "ditto", "mr_mtl")
 
 
 def main() -> None:
@@ -35,9 +38,14 @@ def main() -> None:
         "n_train_volumes": 2, "n_val_volumes": 1, "n_batches_per_epoch": 2,
     }
     fit_cfg = lambda r: {"current_server_round": r, "local_steps": args.local_steps, **cfg}  # noqa: E731
-    if args.algorithm == "fedavg":
+    if args.algorithm in ("fedavg", "fl_nnunet", "central"):
         client_cls = FlexibleNnunetClient
-        strategy = BasicFedAvg(on_fit_config_fn=fit_cfg)
+        if args.algorithm == "central":
+            args.n_clients = 1
+            strategy = BasicFedAvg(on_fit_config_fn=fit_cfg, min_fit_clients=1,
+                                   min_evaluate_clients=1, min_available_clients=1)
+        else:
+            strategy = BasicFedAvg(on_fit_config_fn=fit_cfg)
     else:
         client_cls = make_it_personal(FlexibleNnunetClient, mode=args.algorithm)
         strategy = FedAvgWithAdaptiveConstraint(

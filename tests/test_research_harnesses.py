@@ -71,6 +71,8 @@ def test_evaluate_on_test_helper():
         ("research.flamby.run_experiment", ["--task", "fed_isic2019", "--algorithm", "fenda"]),
         ("research.flamby.run_experiment", ["--task", "fed_ixi", "--algorithm", "apfl"]),
         ("research.picai.run_experiment", ["--algorithm", "mr_mtl"]),
+        ("research.picai.run_experiment", ["--algorithm", "fl_nnunet"]),
+        ("research.picai.run_experiment", ["--algorithm", "central"]),
     ],
 )
 def test_new_research_harnesses_run(module, extra):
