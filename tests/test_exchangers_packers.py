@@ -88,3 +88,74 @@ def test_sparse_coo_packer_roundtrip():
     rec = torch.zeros(4, 5)
     rec[info["indices"][0][0], info["indices"][0][1]] = info["values"][0]
     assert torch.allclose(rec, dense)
+
+
+def test_layer_selection_by_threshold_and_percentage():
+    """Dynamic-exchange selection criteria (reference
+    parameter_selection_criteria): drift-threshold picks only moved layers;
+    percentage picks the top-p% by NORMALIZED drift."""
+    import copy
+
+    import torch.nn as nn
+
+    from fl4health_amd.parameter_exchange.parameter_selection_criteria import (
+        select_layers_by_percentage,
+        select_layers_by_threshold,
+    )
+
+    model = nn.Sequential(nn.Linear(4, 4), nn.Linear(4, 4))
+    init = copy.deepcopy(model)
+    with torch.no_grad():
+        model[0].weight += 10.0  # only layer-0 weight drifts
+    names, n = select_layers_by_threshold(1.0, None, model, init)
+    assert names == ["0.weight"] and n == 1.0
+    names_all, _ = select_layers_by_threshold(-1.0, None, model, init)
+    assert set(names_all) == {"0.weight", "0.bias", "1.weight", "1.bias"}
+    top, k = select_layers_by_percentage(0.25, model, init)
+    assert top == ["0.weight"] and k == 1.0  # 25% of 4 layers = the mover
+
+
+def test_magnitude_score_functions_order():
+    import copy
+
+    import torch.nn as nn
+
+    from fl4health_amd.parameter_exchange.parameter_selection_criteria import (
+        largest_final_magnitude_scores,
+        largest_increase_in_magnitude_scores,
+        largest_magnitude_change_scores,
+        smallest_final_magnitude_scores,
+    )
+
+    model = nn.Linear(3, 1, bias=False)
+    init = copy.deepcopy(model)
+    with torch.no_grad():
+        model.weight.copy_(torch.tensor([[3.0, -1.0, 0.5]]))
+        init.weight.copy_(torch.tensor([[1.0, -2.0, 0.5]]))
+    big = largest_final_magnitude_scores(model, None)["weight"]
+    assert torch.equal(big, torch.tensor([[3.0, 1.0, 0.5]]))
+    small = smallest_final_magnitude_scores(model, None)["weight"]
+    assert float(small.argmax()) == 2  # smallest |w| (0.5) scores highest
+    change = largest_magnitude_change_scores(model, init)["weight"]
+    assert torch.equal(change, torch.tensor([[2.0, 1.0, 0.0]]))
+    inc = largest_increase_in_magnitude_scores(model, init)["weight"]
+    assert float(inc[0, 0]) > float(inc[0, 1])  # |3|-|1| > |-1|-|-2|
+
+
+def test_moon_old_model_buffer_rotation():
+    """MOON keeps at most len_old_models_buffer frozen snapshots, oldest out
+    (reference moon_client buffer semantics)."""
+    import torch.nn as nn
+
+    from fl4health_amd.clients.moon_client import MoonClient
+
+    c = MoonClient(device="cpu", len_old_models_buffer=2)
+    c.model = nn.Linear(2, 2)
+    for r in range(4):
+        with torch.no_grad():
+            c.model.weight += 1.0
+        c.update_after_train(1, {}, {})
+    assert len(c.old_models_list) == 2
+    # newest snapshot reflects the latest weights; all snapshots frozen
+    assert torch.allclose(next(c.old_models_list[-1].parameters()), c.model.weight)
+    assert all(not p.requires_grad for m in c.old_models_list for p in m.parameters())
