@@ -24,18 +24,25 @@ from fl4health_amd.clients.apfl_client import ApflClient
 from fl4health_amd.clients.adaptive_drift_constraint_client import FedProxClient
 from fl4health_amd.clients.basic_client import BasicClient
 from fl4health_amd.clients.ditto_client import DittoClient
+from fl4health_amd.clients.fedper_client import FedPerClient
 from fl4health_amd.clients.fenda_client import FendaClient
+from fl4health_amd.clients.moon_client import MoonClient
+from fl4health_amd.clients.perfcl_client import PerFclClient
 from fl4health_amd.clients.scaffold_client import ScaffoldClient
 from fl4health_amd.common import Parameters
 from fl4health_amd.metrics.metrics import Accuracy, BinarySoftDiceCoefficient
 from fl4health_amd.model_bases.apfl_base import ApflModule
 from fl4health_amd.model_bases.fenda_base import FendaModel
+from fl4health_amd.model_bases.moon_base import MoonModel
+from fl4health_amd.model_bases.perfcl_base import PerFclModel
+from fl4health_amd.model_bases.sequential_split_models import SequentiallySplitExchangeBaseModel
 from fl4health_amd.model_bases.parallel_split_models import ParallelFeatureJoinMode, ParallelSplitHeadModule
 from fl4health_amd.optimizers import FlatProxSGD, FlatScaffoldSGD
 from fl4health_amd.parameter_exchange.flat import FlatParameterView
 from fl4health_amd.servers.base_server import FlServer
 from fl4health_amd.strategies.basic_fedavg import BasicFedAvg
 from fl4health_amd.strategies.fedavg_dynamic_layer import FedAvgDynamicLayer
+from fl4health_amd.strategies.fedopt import FedAdam
 from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithAdaptiveConstraint
 from fl4health_amd.strategies.scaffold import Scaffold
 from research.common import research_argparser, run_and_record
@@ -46,7 +53,10 @@ TASKS = {
     "fed_isic2019": ((3, 64, 64), 8, 6),
     "fed_ixi": ((1, 24, 24, 24), 2, 3),
 }
-ALGORITHMS = ("fedavg", "fedprox", "scaffold", "ditto", "apfl", "fenda")
+ALGORITHMS = (
+    "fedavg", "fedprox", "scaffold", "ditto", "apfl", "fenda",
+    "central", "local", "fedadam", "fedper", "moon", "perfcl",
+)
 
 
 def make_model(task: str) -> nn.Module:
@@ -105,6 +115,8 @@ def build(args, device: str):
     base_cls = {
         "fedavg": BasicClient, "fedprox": FedProxClient, "scaffold": ScaffoldClient,
         "ditto": DittoClient, "apfl": ApflClient, "fenda": FendaClient,
+        "central": BasicClient, "local": BasicClient, "fedadam": BasicClient,
+        "fedper": FedPerClient, "moon": MoonClient, "perfcl": PerFclClient,
     }[args.algorithm]
 
     class Client(base_cls):
@@ -115,15 +127,35 @@ def build(args, device: str):
         def get_model(self, config):
             if args.algorithm == "apfl":
                 return ApflModule(make_model(task), adaptive_alpha=True)
-            if args.algorithm == "fenda":
+            if args.algorithm in ("fenda", "perfcl"):
                 trunk = make_model(task)
                 feat = nn.Sequential(*list(trunk.children())[:-1])
                 feat2 = nn.Sequential(*list(make_model(task).children())[:-1])
                 feat_dim = list(trunk.children())[-1].in_features
+                if args.algorithm == "perfcl":
+                    return PerFclModel(feat, feat2, _FendaHead(feat_dim, ncls))
                 return FendaModel(feat, feat2, _FendaHead(feat_dim, ncls))
+            if args.algorithm in ("fedper", "moon"):
+                trunk = make_model(task)
+                feat = nn.Sequential(*list(trunk.children())[:-1], nn.Flatten(1))
+                head = nn.Linear(list(trunk.children())[-1].in_features, ncls)
+                if args.algorithm == "moon":
+                    return MoonModel(feat, head)
+                return SequentiallySplitExchangeBaseModel(feat, head)
             return make_model(task)
 
         def get_data_loaders(self, config):
+            if args.algorithm == "central":
+                loaders = [
+                    load_site(task, args.data_dir, i, args.batch_size, args.n_train, args.seed)
+                    for i in range(TASKS[task][2])
+                ]
+                train = torch.utils.data.ConcatDataset([dl[0].dataset for dl in loaders])
+                val = torch.utils.data.ConcatDataset([dl[1].dataset for dl in loaders])
+                return (
+                    DataLoader(train, batch_size=args.batch_size, shuffle=True),
+                    DataLoader(val, batch_size=args.batch_size),
+                )
             return load_site(task, args.data_dir, self.site, args.batch_size, args.n_train, args.seed)
 
         def get_criterion(self, config):
@@ -146,7 +178,8 @@ def build(args, device: str):
             if args.algorithm == "ditto" and self.optimizers.get("global") is None:
                 self.optimizers["global"] = FlatProxSGD(self.global_flat_view, lr=args.lr)
 
-    clients = [Client(i, metrics=[Accuracy()], device=device) for i in range(args.n_clients)]
+    n = 1 if args.algorithm in ("central", "local") else args.n_clients
+    clients = [Client(i, metrics=[Accuracy()], device=device) for i in range(n)]
     fit_cfg = lambda r: {"current_server_round": r, "local_steps": args.local_steps}  # noqa: E731
 
     def init_params(model_fn):
@@ -159,8 +192,13 @@ def build(args, device: str):
         )
     elif args.algorithm == "scaffold":
         strategy = Scaffold(initial_parameters=init_params(make_model(task)), on_fit_config_fn=fit_cfg)
-    elif args.algorithm == "fenda":
+    elif args.algorithm in ("fenda", "perfcl", "fedper"):
         strategy = FedAvgDynamicLayer(on_fit_config_fn=fit_cfg)
+    elif args.algorithm == "fedadam":
+        strategy = FedAdam(initial_parameters=init_params(make_model(task)), on_fit_config_fn=fit_cfg)
+    elif args.algorithm in ("central", "local"):
+        strategy = BasicFedAvg(on_fit_config_fn=fit_cfg, min_fit_clients=1,
+                               min_evaluate_clients=1, min_available_clients=1)
     else:
         strategy = BasicFedAvg(on_fit_config_fn=fit_cfg)
     server = FlServer(

@@ -70,6 +70,10 @@ def test_evaluate_on_test_helper():
         ("research.flamby.run_experiment", ["--task", "fed_heart_disease", "--algorithm", "scaffold"]),
         ("research.flamby.run_experiment", ["--task", "fed_isic2019", "--algorithm", "fenda"]),
         ("research.flamby.run_experiment", ["--task", "fed_ixi", "--algorithm", "apfl"]),
+        ("research.flamby.run_experiment", ["--task", "fed_heart_disease", "--algorithm", "moon"]),
+        ("research.flamby.run_experiment", ["--task", "fed_isic2019", "--algorithm", "perfcl"]),
+        ("research.flamby.run_experiment", ["--task", "fed_heart_disease", "--algorithm", "fedadam"]),
+        ("research.flamby.run_experiment", ["--task", "fed_heart_disease", "--algorithm", "central"]),
         ("research.picai.run_experiment", ["--algorithm", "mr_mtl"]),
         ("research.picai.run_experiment", ["--algorithm", "fl_nnunet"]),
         ("research.picai.run_experiment", ["--algorithm", "central"]),
