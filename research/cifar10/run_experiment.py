@@ -27,7 +27,7 @@ from fl4health_amd.strategies.fedavg_with_adaptive_constraint import FedAvgWithA
 from fl4health_amd.strategies.feddg_ga import FedDgGa
 from research.common import research_argparser, run_and_record
 
-ALGORITHMS = ("fedavg", "fedprox", "ditto", "mr_mtl", "ditto_mkmmd", "mr_mtl_mkmmd", "fed_dgga")
+ALGORITHMS = ("fedavg", "fedprox", "ditto", "mr_mtl", "ditto_mkmmd", "mr_mtl_mkmmd", "fed_dgga", "fenda_ditto")
 
 
 def _loaders(args, seed):
@@ -97,6 +97,54 @@ def build(args, device: str):
     elif args.algorithm == "mr_mtl_mkmmd":
         clients = [type("C", (_DataMixin, MrMtlMkMmdClient), {})(i, args, **kw, **mmd_kw) for i in range(args.n_clients)]
         strategy = FedAvgWithAdaptiveConstraint(initial_parameters=init, initial_loss_weight=args.mu, on_fit_config_fn=fit_cfg)
+    elif args.algorithm == "fenda_ditto":
+        import torch.nn as nn
+
+        from fl4health_amd.clients.fenda_ditto_client import FendaDittoClient
+        from fl4health_amd.model_bases.fenda_base import FendaModel
+        from fl4health_amd.model_bases.parallel_split_models import (
+            ParallelFeatureJoinMode,
+            ParallelSplitHeadModule,
+        )
+        from fl4health_amd.model_bases.sequential_split_models import SequentiallySplitExchangeBaseModel
+
+        FEAT = 64 * 8 * 8
+
+        def _extractor():
+            return nn.Sequential(
+                nn.Conv2d(3, 32, 5, padding=2), nn.ReLU(), nn.MaxPool2d(2, 2),
+                nn.Conv2d(32, 64, 5, padding=2), nn.ReLU(), nn.MaxPool2d(2, 2),
+                nn.Flatten(),
+            )
+
+        class _FDHead(ParallelSplitHeadModule):
+            def __init__(self):
+                super().__init__(ParallelFeatureJoinMode.CONCATENATE)
+                self.fc = nn.Linear(2 * FEAT, 10)
+
+            def parallel_output_join(self, local_tensor, global_tensor):
+                return torch.cat([local_tensor.flatten(1), global_tensor.flatten(1)], dim=1)
+
+            def head_forward(self, x):
+                return self.fc(x)
+
+        class _FDMixin(_DataMixin, FendaDittoClient):
+            def get_model(self, config):
+                return FendaModel(_extractor(), _extractor(), _FDHead())
+
+            def get_global_model(self, config):
+                return SequentiallySplitExchangeBaseModel(_extractor(), nn.Linear(FEAT, 10))
+
+            def get_optimizer(self, config):
+                return FlatProxSGD(self.flat_view, lr=self.args.lr)
+
+        init_fd = Parameters(
+            [FlatParameterView(SequentiallySplitExchangeBaseModel(_extractor(), nn.Linear(FEAT, 10))).flat.clone()]
+        )
+        clients = [_FDMixin(i, args, **kw) for i in range(args.n_clients)]
+        strategy = FedAvgWithAdaptiveConstraint(
+            initial_parameters=init_fd, initial_loss_weight=args.mu, on_fit_config_fn=fit_cfg
+        )
     elif args.algorithm == "fed_dgga":
         clients = [type("C", (_TorchOptMixin,), {})(i, args, **kw) for i in range(args.n_clients)]
         strategy = FedDgGa(on_fit_config_fn=fit_cfg)

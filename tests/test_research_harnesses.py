@@ -37,6 +37,12 @@ def test_cifar10_harness_fedprox(tmp_path):
     assert len(list(tmp_path.glob("*.json"))) == 1
 
 
+def test_cifar10_harness_fenda_ditto():
+    out = _run("research.cifar10.run_experiment", "--algorithm", "fenda_ditto")
+    rec = json.loads(out.strip().splitlines()[-1])
+    assert rec["final_loss"] is not None
+
+
 def test_ag_news_harness_dynamic_layer():
     out = _run("research.ag_news.run_experiment", "--algorithm", "dynamic_layer", "--lr", "1e-4")
     rec = json.loads(out.strip().splitlines()[-1])
