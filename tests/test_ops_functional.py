@@ -248,3 +248,23 @@ def test_conv3d_depth_decomposition_matches_conv3d_cpu():
     assert type(m[0]) is Cdna3dConv
     y = m(torch.randn(1, 2, 4, 4, 4))  # CPU falls back to F.conv3d
     assert y.shape == (1, 2, 4, 4, 4)
+
+
+def test_bn_v2_geometry_invariants():
+    """bn_geom (bn_ops.hip) contract mirrored on CPU: PX is a pow2 covering
+    C/2 capped at 128, PX*PY == 1024, and the (PX,PY) grid covers every
+    channel pair for the ResNet channel family."""
+    def bn_geom(C):
+        half = C >> 1
+        p = 8
+        while p < half and p < 128:
+            p <<= 1
+        return p, 1024 // p
+
+    for C in (16, 32, 64, 96, 128, 256, 512):
+        px, py = bn_geom(C)
+        assert px * py == 1024
+        assert px & (px - 1) == 0  # pow2 (exact tree reduce over py)
+        cblocks = (C // 2 + px - 1) // px
+        assert cblocks * px >= C // 2  # every pair slot covered
+        assert px <= 128
