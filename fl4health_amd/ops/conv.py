@@ -1,12 +1,13 @@
 """CdnaConv2d: 3x3 stride-1 NHWC bf16 convolution on the hand-written MFMA
 kernel (ops/csrc/conv_ops.hip), with full autograd:
 
-- forward: direct MFMA kernel (beats MIOpen on the mid-depth ResNet shapes,
-  profiles/kernels_summary.md)
-- backward data: the SAME forward kernel with flipped-tap, C/K-transposed
+- forward: direct MFMA kernel family (kb32 / kzloop variants; adopted per
+  shape where they beat TUNED MIOpen, profiles/kernels_summary.md)
+- backward data: the SAME forward kernels with flipped-tap, C/K-transposed
   weights (dx = conv(dy, rot(W)) — math verified against torch autograd)
-- backward weights: 9 shifted GEMMs on rocBLAS (dW_tap = X_shift^T @ dY, the
-  natural NHWC contraction — library MFMA; no im2col materialization)
+- backward weights: MIOpen's tuned wrw igemm by default; the hand-written
+  transposed-LDS 9-tap-reuse kernel (conv3x3_wrw) where it measures faster
+  (layer-1 family, 1.23x — see _WrwConv2dFn)
 
 The module keeps the standard nn.Conv2d parameter layout ([K, C, 3, 3] fp32)
 so class-swapping via `convert_conv3x3_to_cdna` stays state_dict- and
