@@ -127,3 +127,46 @@ def test_rdp_accountant_monotonicity(steps, z):
     e3 = acc.get_epsilon(0.01, z + 0.5, steps, delta=1e-5)
     assert e2 >= e1 - 1e-9
     assert e3 <= e1 + 1e-9
+
+
+@given(
+    tp=st.lists(st.integers(0, 50), min_size=1, max_size=12),
+    fp=st.integers(0, 50),
+    fn=st.integers(0, 50),
+)
+@settings(max_examples=40, deadline=None)
+def test_dice_from_counts_bounded_and_monotone(tp, fp, fn):
+    """Dice scores live in [0, 1]; adding true positives never lowers the
+    score of an entry (holding FP/FN fixed)."""
+    import torch
+
+    from fl4health_amd.metrics.metrics_utils import compute_dice_on_count_tensors
+
+    tp_t = torch.tensor([float(v) for v in tp])
+    fp_t = torch.full_like(tp_t, float(fp))
+    fn_t = torch.full_like(tp_t, float(fn))
+    d = compute_dice_on_count_tensors(tp_t, fp_t, fn_t, zero_division=0.0)
+    assert ((d >= 0) & (d <= 1)).all()
+    d2 = compute_dice_on_count_tensors(tp_t + 1, fp_t, fn_t, zero_division=0.0)
+    assert (d2 >= d - 1e-6).all()
+
+
+@given(sizes=st.lists(st.integers(9, 40), min_size=3, max_size=3),
+       max_levels=st.integers(2, 5))
+@settings(max_examples=30, deadline=None)
+def test_nnunet_plan_patch_always_network_compatible(sizes, max_levels):
+    """Property form of the divisibility regression: ANY volume geometry must
+    plan a patch every dim of which divides 2^n_stages (UNet3D pools after
+    every encoder level)."""
+    from fl4health_amd.preprocessing.nnunet import plan_experiment
+
+    fp = {
+        "shapes_after_crop": [sizes],
+        "spacings": [[1.0, 1.0, 1.0]],
+        "foreground_intensity_properties_per_channel": {"0": {"mean": 0.0, "std": 1.0}},
+    }
+    dj = {"numTraining": 1, "channel_names": {"0": "c"}, "labels": {"background": 0, "fg": 1}}
+    plans = plan_experiment(fp, dj, max_patch_voxels=24 ** 3, max_levels=max_levels)
+    cfg = plans["configurations"]["3d_fullres"]
+    div = 2 ** cfg["n_stages"]
+    assert all(p % div == 0 and p >= div for p in cfg["patch_size"])
