@@ -596,3 +596,45 @@ def test_wandb_reporter_with_mock():
     assert calls["log"][0] == {"loss": 1.5, "acc": 0.2, "fl_round": 3}
     assert calls["log"][1] == {"val": 7, "fl_round": 4, "step": 12}
     assert calls["finish"] == 1
+
+
+def test_snapshotters_roundtrip_module_optimizer_scheduler():
+    """Snapshot/restore parity for the three stateful client attributes
+    (reference tests/utils/snapshotter_test.py behaviors)."""
+    import torch.nn as nn
+
+    from fl4health_amd.utils.snapshotter import (
+        LRSchedulerSnapshotter,
+        NumberSnapshotter,
+        OptimizerSnapshotter,
+        TorchModuleSnapshotter,
+    )
+
+    torch.manual_seed(0)
+    model = nn.Linear(4, 2)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    sched = torch.optim.lr_scheduler.StepLR(opt, step_size=2, gamma=0.5)
+    # create optimizer/scheduler state
+    for _ in range(3):
+        loss = model(torch.randn(8, 4)).sum()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        sched.step()
+    m_snap = TorchModuleSnapshotter().save(model)
+    o_snap = OptimizerSnapshotter().save(opt)
+    s_snap = LRSchedulerSnapshotter().save(sched)
+    n_snap = NumberSnapshotter().save(7)
+
+    model2 = nn.Linear(4, 2)
+    opt2 = torch.optim.SGD(model2.parameters(), lr=0.1, momentum=0.9)
+    sched2 = torch.optim.lr_scheduler.StepLR(opt2, step_size=2, gamma=0.5)
+    TorchModuleSnapshotter().load(model2, m_snap)
+    OptimizerSnapshotter().load(opt2, o_snap)
+    LRSchedulerSnapshotter().load(sched2, s_snap)
+    assert all(torch.equal(a, b) for a, b in zip(model.parameters(), model2.parameters()))
+    # momentum buffers restored
+    buf = opt2.state[model2.weight].get("momentum_buffer")
+    assert buf is not None and torch.equal(buf, opt.state[model.weight]["momentum_buffer"])
+    assert sched2.last_epoch == sched.last_epoch and sched2.get_last_lr() == sched.get_last_lr()
+    assert NumberSnapshotter().load(0, n_snap) == 7
