@@ -246,6 +246,7 @@ class DistributedRuntime:
             if payload is None or cid not in proxies:
                 continue
             if isinstance(payload, dict) and "error" in payload:
+                log.error("rank %s evaluate failed: %s", cid, payload["error"])
                 failures.append(RuntimeError(payload["error"]))
             else:
                 loss, n, metrics = payload
@@ -552,6 +553,7 @@ class DistributedRuntime:
             if h is None:
                 continue
             if "error" in h:
+                log.error("rank %s fit failed: %s", cid, h["error"])
                 failures.append(RuntimeError(h["error"]))
                 continue
             tensors: list[torch.Tensor | None] = [None] * h["n_tensors"]

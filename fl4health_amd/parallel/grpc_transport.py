@@ -157,6 +157,8 @@ class GrpcServerTransport:
                 try:
                     results.append((proxy, fut.result()))
                 except Exception as e:  # noqa: BLE001 — per-client failure policy
+                    log.exception("grpc client %s call failed (accept_failures=%s)",
+                                  proxy.cid, self.accept_failures)
                     if not self.accept_failures:
                         raise
                     failures.append(e)
