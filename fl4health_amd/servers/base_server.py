@@ -172,6 +172,11 @@ class FlServer:
         results, failures = self.fit_clients(instructions, timeout)
         if failures and not self.accept_failures:
             self._terminate_after_unacceptable_failures(failures)
+        if not results:
+            log.warning(
+                "fit round %d produced NO results (%d failures accepted)",
+                server_round, len(failures),
+            )
         params, metrics = self.aggregate_fit(server_round, results, failures)
         if params is not None:
             self.parameters = params
@@ -213,6 +218,13 @@ class FlServer:
         results, failures = self.evaluate_clients(instructions, timeout)
         if failures and not self.accept_failures:
             self._terminate_after_unacceptable_failures(failures)
+        if not results:
+            # every client failed (accept_failures swallowed them): say so
+            # loudly instead of quietly recording an empty history round
+            log.warning(
+                "evaluate round %d produced NO results (%d failures accepted)",
+                server_round, len(failures),
+            )
         loss_aggregated, metrics_aggregated = self.strategy.aggregate_evaluate(server_round, results, failures)
         val_metrics, test_metrics = self._unpack_metrics(metrics_aggregated)
         eval_end = datetime.datetime.now()
