@@ -38,15 +38,19 @@ class ApflModule(PartialLayerExchangeModel):
 
     def update_alpha(self) -> None:
         """Closed-form alpha gradient from current grads (reference :83-117):
-        grad_alpha = sum_p <grad(local_p or global_p), local_p - global_p>."""
-        grad_alpha = 0.0
+        grad_alpha = sum_p <grad(local_p or global_p), local_p - global_p>.
+
+        The dot products accumulate ON DEVICE (K16); a per-parameter float()
+        here would be one host sync per layer per step."""
+        acc = None
         for lp, gp in zip(self.local_model.parameters(), self.global_model.parameters()):
             diff = (lp.detach() - gp.detach()).reshape(-1)
             grad = lp.grad if lp.grad is not None else gp.grad
             if grad is None:
                 continue
-            grad_alpha += float(torch.dot(grad.detach().reshape(-1), diff))
-        grad_alpha += 0.02 * self.alpha  # l2 regularization on alpha (reference :107)
+            d = torch.dot(grad.detach().reshape(-1), diff)
+            acc = d if acc is None else acc + d
+        grad_alpha = (float(acc) if acc is not None else 0.0) + 0.02 * self.alpha  # l2 reg (reference :107)
         alpha = self.alpha - self.alpha_lr * grad_alpha
         self.alpha = max(0.0, min(1.0, alpha))
 
