@@ -67,3 +67,44 @@ def test_scaffold_sgd_step_applies_variate_correction():
     opt.step()
     expect = w0 - 0.1 * (g + c - ci)
     assert torch.allclose(fv.params_region, expect, atol=1e-5)
+
+
+def test_evaluate_after_fit_packs_val_metrics():
+    """fit() with evaluate_after_fit runs validation inside the fit round and
+    merges val metrics (plus the packed checkpoint loss) into the fit metrics
+    (reference basic_client evaluate-after-fit contract)."""
+    import torch.nn as nn
+    from torch.utils.data import DataLoader, TensorDataset
+
+    from fl4health_amd.clients.basic_client import BasicClient
+    from fl4health_amd.common import Parameters
+    from fl4health_amd.metrics.metrics import Accuracy
+    from fl4health_amd.parameter_exchange.flat import FlatParameterView
+
+    class C(BasicClient):
+        def get_model(self, config):
+            return nn.Linear(4, 2)
+
+        def get_data_loaders(self, config):
+            ds = TensorDataset(torch.randn(64, 4), torch.randint(0, 2, (64,)))
+            return DataLoader(ds, batch_size=16), DataLoader(ds, batch_size=16)
+
+        def get_optimizer(self, config):
+            return torch.optim.SGD(self.model.parameters(), lr=0.01)
+
+        def get_criterion(self, config):
+            return nn.CrossEntropyLoss()
+
+    torch.manual_seed(0)
+    c = C(device="cpu", metrics=[Accuracy()])
+    params = Parameters([FlatParameterView(nn.Linear(4, 2)).flat.clone()])
+    cfg = {"current_server_round": 1, "local_steps": 2,
+           "evaluate_after_fit": True, "pack_losses_with_val_metrics": True}
+    _, n, metrics = c.fit(params, cfg)
+    assert n == 64
+    assert any(k.startswith("val") for k in metrics), metrics
+    assert "val - checkpoint" in metrics and metrics["val - checkpoint"] > 0
+    # without the flag, fit metrics stay train-only
+    c2 = C(device="cpu", metrics=[Accuracy()])
+    _, _, m2 = c2.fit(params, {"current_server_round": 1, "local_steps": 2})
+    assert not any(k.startswith("val") for k in m2)
