@@ -65,3 +65,37 @@ def test_grpc_star_topology_round_trip(tmp_path):
         c_out, c_err = c.communicate(timeout=60)
         assert c.returncode == 0, c_err[-2000:]
         assert f"CLIENT_DONE {i}" in c_out
+
+
+def test_docker_example_entrypoints_end_to_end(tmp_path):
+    """The containerized server/client entry points run a full FL session
+    over localhost gRPC (what docker-compose launches across containers)."""
+    import os
+    import subprocess
+    import sys
+    import time
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parents[1]
+    env = dict(os.environ, PYTHONPATH=str(root))
+    srv = subprocess.Popen(
+        [sys.executable, "-m", "examples.docker_basic_example.server",
+         "--address", "127.0.0.1:18113", "--rounds", "1", "--local_steps", "2"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, env=env, cwd=str(root),
+    )
+    time.sleep(2.0)
+    clients = [
+        subprocess.Popen(
+            [sys.executable, "-m", "examples.docker_basic_example.client",
+             "--server", "127.0.0.1:18113", "--seed", str(i)],
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL, env=env, cwd=str(root),
+        )
+        for i in range(2)
+    ]
+    try:
+        out, _ = srv.communicate(timeout=240)
+    finally:
+        for c in clients:
+            c.terminate()
+    assert srv.returncode == 0, out[-2000:]
+    assert "losses:" in out
