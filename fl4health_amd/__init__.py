@@ -7,4 +7,4 @@ collectives over xGMI (one client process per GPU) in place of the
 reference's Flower gRPC NumPy round-trip.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

@@ -14,7 +14,7 @@ CSRC = os.path.join(ROOT, "fl4health_amd", "ops", "csrc")
 
 setup(
     name="fl4health_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=["fl4health_amd"],
     ext_modules=[
         CUDAExtension(
