@@ -694,3 +694,29 @@ def test_fsdp_actually_shards_at_world_two(tmp_path):
     )
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
     assert out.stdout.count("FSDP_OK") == 2
+
+
+def test_bench_script_single_rank_contract(tmp_path):
+    """Driver contract, N=1 direct invocation: one JSON line with every field
+    the driver parses, correct semantics (value = whole-job aggregate,
+    ms_per_step time-like, weak scaling, bf16 declared)."""
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "1", "--steps", "2", "--warmup", "1",
+         "--shard_size", "256", "--batch_size", "32"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(ROOT),
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    import json as _json
+
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-500:]
+    rec = _json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+                "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config"):
+        assert key in rec, key
+    assert rec["n_gpus"] == 1 and rec["steps"] == 2 and rec["warmup"] == 1
+    assert rec["higher_is_better"] is True and rec["scaling"] == "weak"
+    assert rec["dtype"] == "bf16" and "synthetic" in rec["data"]
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
+    assert isinstance(rec["config"], dict) and "model" in rec["config"]
