@@ -717,6 +717,7 @@ def test_bench_script_single_rank_contract(tmp_path):
         assert key in rec, key
     assert rec["n_gpus"] == 1 and rec["steps"] == 2 and rec["warmup"] == 1
     assert rec["higher_is_better"] is True and rec["scaling"] == "weak"
-    assert rec["dtype"] == "bf16" and "synthetic" in rec["data"]
+    # bf16 on GPU; the CPU CI run rightly declares fp32
+    assert rec["dtype"] in ("bf16", "fp32") and "synthetic" in rec["data"]
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
     assert isinstance(rec["config"], dict) and "model" in rec["config"]
