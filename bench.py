@@ -136,6 +136,12 @@ def main() -> None:
     torch.backends.cudnn.benchmark = True  # MIOpen algo find during warmup
 
     has_gpu = torch.cuda.is_available()
+    if has_gpu:
+        # the flagship numbers are only meaningful with the CDNA4 extension:
+        # never silently fall back to eager/MIOpen-only on a GPU box
+        from fl4health_amd.ops import functional as _F
+
+        assert _F.HAS_EXT, "fl4health_amd._C extension missing on a GPU machine - rebuild with setup.py build_ext --inplace"
     runtime = DistributedRuntime(backend="nccl" if has_gpu else "gloo")
     rank, world = runtime.rank, runtime.world_size
     device = runtime.comm_device if has_gpu else torch.device("cpu")
